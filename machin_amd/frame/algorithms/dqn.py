@@ -1,0 +1,318 @@
+"""DQN family: vanilla, fixed-target, double (dueling via model).
+
+Parity target: reference ``machin/frame/algorithms/dqn.py`` (DQN :22):
+same constructor surface, act/criticize/store/update API, three update
+modes (:357-470), epsilon-greedy acting (:253-291), soft/hard target
+updates (:460-467), ``generate_config``/``init_from_config`` (:504+).
+"""
+from typing import Any, Callable, Dict, List, Union
+
+import numpy as np
+import torch as t
+import torch.nn as nn
+
+from ...utils.conf import Config
+from ..buffers.buffer import Buffer
+from ..transition import Transition, TransitionBase
+from .base import TorchFramework
+from .utils import (
+    FakeOptimizer,
+    assert_and_get_valid_criterion,
+    assert_and_get_valid_lr_scheduler,
+    assert_and_get_valid_models,
+    assert_and_get_valid_optimizer,
+    hard_update,
+    safe_call,
+    safe_return,
+    soft_update,
+)
+
+
+class DQN(TorchFramework):
+    """Deep Q-network framework (discrete actions)."""
+
+    _is_top = ["qnet", "qnet_target"]
+    _is_restorable = ["qnet_target"]
+
+    def __init__(
+        self,
+        qnet: nn.Module,
+        qnet_target: nn.Module,
+        optimizer: Callable,
+        criterion: Callable,
+        *_,
+        lr_scheduler: Callable = None,
+        lr_scheduler_args: tuple = None,
+        lr_scheduler_kwargs: tuple = None,
+        batch_size: int = 100,
+        epsilon_decay: float = 0.9999,
+        update_rate: Union[float, None] = 0.005,
+        update_steps: Union[int, None] = None,
+        learning_rate: float = 0.001,
+        discount: float = 0.99,
+        gradient_max: float = np.inf,
+        replay_size: int = 500000,
+        replay_device: Union[str, t.device] = "cpu",
+        replay_buffer: Buffer = None,
+        mode: str = "double",
+        visualize: bool = False,
+        visualize_dir: str = "",
+        **__,
+    ):
+        super().__init__()
+        if mode not in ("vanilla", "fixed_target", "double"):
+            raise ValueError(f"Unknown DQN mode: {mode!r}")
+        self.batch_size = batch_size
+        self.epsilon_decay = epsilon_decay
+        self.update_rate = update_rate
+        self.update_steps = update_steps
+        self.discount = discount
+        self.grad_max = gradient_max
+        self.mode = mode
+        self.visualize = visualize
+        self.visualize_dir = visualize_dir
+        self.epsilon = 1.0
+        self._update_counter = 0
+
+        self.qnet = qnet
+        if mode == "vanilla":
+            self.qnet_target = qnet
+        else:
+            self.qnet_target = qnet_target
+            hard_update(self.qnet_target, self.qnet)
+        if update_rate is not None and update_steps is not None:
+            raise ValueError(
+                "Specify either update_rate (soft) or update_steps (hard), "
+                "not both."
+            )
+
+        if isinstance(optimizer, t.optim.Optimizer):
+            self.qnet_optim = optimizer
+        else:
+            self.qnet_optim = optimizer(self.qnet.parameters(), lr=learning_rate)
+        self.replay_buffer = (
+            Buffer(replay_size, replay_device)
+            if replay_buffer is None
+            else replay_buffer
+        )
+        if lr_scheduler is not None:
+            args = (lr_scheduler_args or ((),))[0]
+            kwargs = (lr_scheduler_kwargs or ({},))[0]
+            self.qnet_lr_sch = lr_scheduler(self.qnet_optim, *args, **kwargs)
+        else:
+            self.qnet_lr_sch = None
+        self.criterion = criterion() if isinstance(criterion, type) else criterion
+
+    # ------------------------------------------------------------------
+    @property
+    def optimizers(self):
+        return [self.qnet_optim]
+
+    @optimizers.setter
+    def optimizers(self, optimizers):
+        self.qnet_optim = optimizers[0]
+
+    @property
+    def lr_schedulers(self):
+        return [self.qnet_lr_sch] if self.qnet_lr_sch is not None else []
+
+    # ------------------------------------------------------------------
+    # acting
+    # ------------------------------------------------------------------
+    def act_discrete(self, state: Dict[str, Any], use_target: bool = False, **__):
+        """Greedy action as an int64 tensor of shape [batch, 1]."""
+        q = self._criticize(state, use_target)
+        return t.argmax(q, dim=1).view(-1, 1)
+
+    def act_discrete_with_noise(
+        self,
+        state: Dict[str, Any],
+        use_target: bool = False,
+        decay_epsilon: bool = True,
+        **__,
+    ):
+        """Epsilon-greedy action; epsilon multiplies by
+        ``epsilon_decay`` per call when ``decay_epsilon``."""
+        q = self._criticize(state, use_target)
+        batch, n_actions = q.shape[0], q.shape[1]
+        greedy = t.argmax(q, dim=1).view(-1, 1)
+        if t.rand(1).item() < self.epsilon:
+            result = t.randint(0, n_actions, (batch, 1))
+        else:
+            result = greedy
+        if decay_epsilon:
+            self.epsilon *= self.epsilon_decay
+        return result
+
+    def _criticize(self, state: Dict[str, Any], use_target: bool = False, **__):
+        net = self.qnet_target if use_target else self.qnet
+        return safe_return(safe_call(net, state))
+
+    # ------------------------------------------------------------------
+    # storing
+    # ------------------------------------------------------------------
+    def store_transition(self, transition: Union[Transition, Dict]):
+        self.replay_buffer.store_episode(
+            [transition],
+            required_attrs=("state", "action", "next_state", "reward", "terminal"),
+        )
+
+    def store_episode(self, episode: List[Union[Transition, Dict]]):
+        self.replay_buffer.store_episode(
+            episode,
+            required_attrs=("state", "action", "next_state", "reward", "terminal"),
+        )
+
+    # ------------------------------------------------------------------
+    # updating
+    # ------------------------------------------------------------------
+    def update(
+        self, update_value=True, update_target=True, concatenate_samples=True, **__
+    ):
+        """One gradient step from a replay sample.
+
+        Returns the mean value loss (float)."""
+        (
+            batch_size,
+            (state, action, reward, next_state, terminal, others),
+        ) = self.replay_buffer.sample_batch(
+            self.batch_size,
+            concatenate_samples,
+            sample_method="random_unique",
+            sample_attrs=["state", "action", "reward", "next_state", "terminal", "*"],
+        )
+        if batch_size == 0:
+            return 0.0
+        self.qnet.train()
+
+        with t.no_grad():
+            if self.mode == "vanilla":
+                q_next = self._criticize(next_state)
+                target = q_next.max(dim=1, keepdim=True)[0]
+            elif self.mode == "fixed_target":
+                q_next = self._criticize(next_state, use_target=True)
+                target = q_next.max(dim=1, keepdim=True)[0]
+            else:  # double
+                online_next = self._criticize(next_state)
+                best_action = online_next.argmax(dim=1, keepdim=True)
+                q_next = self._criticize(next_state, use_target=True)
+                target = q_next.gather(dim=1, index=best_action.to(q_next.device))
+            device = target.device
+            reward = reward.to(device).float().view(batch_size, 1)
+            terminal = terminal.to(device).float().view(batch_size, 1)
+            y = reward + self.discount * (1.0 - terminal) * target
+
+        q = self._criticize(state)
+        action_index = self._sampled_action_index(action, q)
+        q_taken = q.gather(dim=1, index=action_index)
+        loss = self.criterion(q_taken, y.to(q_taken.dtype))
+
+        if self.visualize:
+            self.visualize_model(loss, "qnet", self.visualize_dir)
+
+        if update_value:
+            self.qnet_optim.zero_grad(set_to_none=True)
+            self._backward(loss)
+            nn.utils.clip_grad_norm_(self.qnet.parameters(), self.grad_max)
+            self.qnet_optim.step()
+
+        if update_target and self.mode != "vanilla":
+            if self.update_rate is not None:
+                soft_update(self.qnet_target, self.qnet, self.update_rate)
+            else:
+                self._update_counter += 1
+                if self._update_counter % self.update_steps == 0:
+                    hard_update(self.qnet_target, self.qnet)
+        return float(loss.detach().item())
+
+    @staticmethod
+    def _sampled_action_index(action: Dict[str, t.Tensor], q: t.Tensor):
+        idx = action["action"]
+        return idx.to(device=q.device, dtype=t.long).view(-1, 1)
+
+    def update_lr_scheduler(self):
+        if self.qnet_lr_sch is not None:
+            self.qnet_lr_sch.step()
+
+    # ------------------------------------------------------------------
+    def load(self, model_dir, network_map=None, version=-1):
+        super().load(model_dir, network_map, version)
+        with t.no_grad():
+            hard_update(self.qnet, self.qnet_target)
+
+    # ------------------------------------------------------------------
+    @classmethod
+    def generate_config(cls, config: Union[Dict[str, Any], Config]):
+        default = {
+            "frame": cls.__name__,
+            "models": ["QNet", "QNet"],
+            "model_args": ((), ()),
+            "model_kwargs": ({}, {}),
+            "optimizer": "Adam",
+            "criterion": "MSELoss",
+            "criterion_args": (),
+            "criterion_kwargs": {},
+            "lr_scheduler": None,
+            "lr_scheduler_args": None,
+            "lr_scheduler_kwargs": None,
+            "batch_size": 100,
+            "epsilon_decay": 0.9999,
+            "update_rate": 0.005,
+            "update_steps": None,
+            "learning_rate": 0.001,
+            "discount": 0.99,
+            "gradient_max": 1e9,
+            "replay_size": 500000,
+            "replay_device": "cpu",
+            "mode": "double",
+            "visualize": False,
+            "visualize_dir": "",
+        }
+        config = config or {}
+        data = config.data if isinstance(config, Config) else dict(config)
+        frame_config = dict(default)
+        frame_config.update(data.get("frame_config", {}))
+        data["frame"] = frame_config["frame"]
+        data["frame_config"] = frame_config
+        return Config(**data)
+
+    @classmethod
+    def init_from_config(
+        cls, config: Union[Dict[str, Any], Config], model_device="cpu"
+    ):
+        data = config.data if isinstance(config, Config) else dict(config)
+        fc = data["frame_config"]
+        model_cls = assert_and_get_valid_models(fc["models"])
+        model_args = fc.get("model_args", ((), ()))
+        model_kwargs = fc.get("model_kwargs", ({}, {}))
+        models = [
+            m(*args, **kwargs).to(model_device)
+            for m, args, kwargs in zip(model_cls, model_args, model_kwargs)
+        ]
+        optimizer = assert_and_get_valid_optimizer(fc["optimizer"])
+        criterion = assert_and_get_valid_criterion(fc["criterion"])(
+            *fc.get("criterion_args", ()), **fc.get("criterion_kwargs", {})
+        )
+        lr_scheduler = assert_and_get_valid_lr_scheduler(fc.get("lr_scheduler"))
+        return cls(
+            *models,
+            optimizer=optimizer,
+            criterion=criterion,
+            lr_scheduler=lr_scheduler,
+            **{
+                k: v
+                for k, v in fc.items()
+                if k
+                not in (
+                    "frame",
+                    "models",
+                    "model_args",
+                    "model_kwargs",
+                    "optimizer",
+                    "criterion",
+                    "criterion_args",
+                    "criterion_kwargs",
+                    "lr_scheduler",
+                )
+            },
+        )

@@ -1,0 +1,286 @@
+"""machin_amd.ops — gfx950 HIP kernels with CPU fallbacks.
+
+Every hot data-path op identified in SURVEY.md §2.6-B has two
+implementations:
+
+* a hand-written CDNA4 HIP kernel (``machin_amd/ops/hip/*.hip``),
+  compiled in-tree into the ``machin_amd.ops._machin_hip`` extension
+  (``setup.py build_ext --inplace`` with ``PYTORCH_ROCM_ARCH=gfx950``);
+* a pure-PyTorch fallback used on CPU.
+
+Dispatch policy: CUDA tensors REQUIRE the extension — if it is missing
+on a GPU machine we raise instead of silently falling back to eager
+torch, so a benchmark can never accidentally measure the fallback.
+"""
+from typing import List, Optional, Sequence
+
+import torch as t
+
+_ext = None
+_ext_error: Optional[str] = None
+
+
+def _load_ext():
+    global _ext, _ext_error
+    if _ext is not None or _ext_error is not None:
+        return _ext
+    try:
+        from . import _machin_hip  # type: ignore
+
+        _ext = _machin_hip
+    except ImportError as e:
+        _ext_error = str(e)
+    return _ext
+
+
+def available() -> bool:
+    """True if the gfx950 HIP extension is importable."""
+    return _load_ext() is not None
+
+
+def _require_ext():
+    ext = _load_ext()
+    if ext is None:
+        raise RuntimeError(
+            "machin_amd HIP extension is not built but a CUDA tensor was "
+            "passed. Build it in-tree with: python setup.py build_ext "
+            f"--inplace  (import error: {_ext_error})"
+        )
+    return ext
+
+
+# ======================================================================
+# fused multi-tensor polyak update (soft_update hot path)
+# ======================================================================
+def polyak_update_(
+    targets: List[t.Tensor], sources: List[t.Tensor], tau: float
+) -> None:
+    """target = target*(1-tau) + source*tau, fused over all tensors.
+
+    Replaces the per-parameter python loop of the reference
+    (machin/frame/algorithms/utils.py:8-27) with one HIP kernel pass
+    over every parameter chunk (HBM-bound: one read+write of target,
+    one read of source).
+    """
+    if not targets:
+        return
+    if targets[0].is_cuda:
+        ext = _require_ext()
+        ext.multi_tensor_polyak(targets, sources, float(tau))
+    else:
+        t._foreach_mul_(targets, 1.0 - tau)
+        t._foreach_add_(targets, sources, alpha=tau)
+
+
+# ======================================================================
+# reverse-time scans: discounted returns, GAE, n-step, V-trace
+# ======================================================================
+def discounted_returns(
+    rewards: t.Tensor,
+    terminals: t.Tensor,
+    gamma: float,
+    bootstrap: Optional[t.Tensor] = None,
+) -> t.Tensor:
+    """R_t = r_t + gamma * (1-d_t) * R_{t+1}; shape [T] or [T, B].
+
+    Reference semantics: machin/frame/algorithms/a2c.py:275-313 (the
+    lambda==1 GAE branch).
+    """
+    rewards = rewards.float()
+    squeeze = rewards.dim() == 1
+    if squeeze:
+        rewards = rewards.unsqueeze(1)
+        terminals = terminals.view(-1, 1)
+        if bootstrap is not None:
+            bootstrap = bootstrap.view(1)
+    T, B = rewards.shape
+    terminals = terminals.view(T, B).to(rewards.dtype)
+    if bootstrap is None:
+        bootstrap = t.zeros(B, dtype=rewards.dtype, device=rewards.device)
+    if rewards.is_cuda:
+        ext = _require_ext()
+        out = ext.discounted_returns(
+            rewards.contiguous(), terminals.contiguous(),
+            bootstrap.contiguous().float(), float(gamma),
+        )
+    else:
+        out = t.empty_like(rewards)
+        running = bootstrap.clone().to(rewards.dtype)
+        for i in range(T - 1, -1, -1):
+            running = rewards[i] + gamma * (1.0 - terminals[i]) * running
+            out[i] = running
+    return out.squeeze(1) if squeeze else out
+
+
+def gae(
+    rewards: t.Tensor,
+    values: t.Tensor,
+    next_values: t.Tensor,
+    terminals: t.Tensor,
+    gamma: float,
+    lam: float,
+) -> t.Tensor:
+    """Generalized advantage estimation; shapes [T] or [T, B].
+
+    A_t = delta_t + gamma*lam*(1-d_t)*A_{t+1},
+    delta_t = r_t + gamma*(1-d_t)*V_{t+1} - V_t.
+    Reference semantics: machin/frame/algorithms/a2c.py:269-326.
+    """
+    squeeze = rewards.dim() == 1
+    rewards = rewards.float()
+    if squeeze:
+        rewards = rewards.unsqueeze(1)
+        values = values.view(-1, 1)
+        next_values = next_values.view(-1, 1)
+        terminals = terminals.view(-1, 1)
+    T, B = rewards.shape
+    values = values.view(T, B).float()
+    next_values = next_values.view(T, B).float()
+    terminals = terminals.view(T, B).to(rewards.dtype)
+    if rewards.is_cuda:
+        ext = _require_ext()
+        out = ext.gae(
+            rewards.contiguous(), values.contiguous(),
+            next_values.contiguous(), terminals.contiguous(),
+            float(gamma), float(lam),
+        )
+    else:
+        nd = 1.0 - terminals
+        delta = rewards + gamma * nd * next_values - values
+        out = t.empty_like(delta)
+        running = t.zeros(B, dtype=rewards.dtype, device=rewards.device)
+        for i in range(T - 1, -1, -1):
+            running = delta[i] + gamma * lam * nd[i] * running
+            out[i] = running
+    return out.squeeze(1) if squeeze else out
+
+
+def nstep_returns(
+    rewards: t.Tensor, terminals: t.Tensor, gamma: float, n: int
+) -> t.Tensor:
+    """n-step truncated return per timestep; shape [T] or [T, B].
+
+    G_t = sum_{k=0}^{n-1} gamma^k r_{t+k} (stopping at terminal).
+    Reference semantics: machin/frame/algorithms/rainbow.py:179-189.
+    """
+    squeeze = rewards.dim() == 1
+    rewards = rewards.float()
+    if squeeze:
+        rewards = rewards.unsqueeze(1)
+        terminals = terminals.view(-1, 1)
+    T, B = rewards.shape
+    terminals = terminals.view(T, B).to(t.bool)
+    out = t.zeros_like(rewards)
+    for ti in range(T):
+        g = t.zeros(B, dtype=rewards.dtype, device=rewards.device)
+        factor = t.ones(B, dtype=rewards.dtype, device=rewards.device)
+        alive = t.ones(B, dtype=rewards.dtype, device=rewards.device)
+        for k in range(n):
+            if ti + k >= T:
+                break
+            g = g + factor * alive * rewards[ti + k]
+            alive = alive * (~terminals[ti + k]).to(rewards.dtype)
+            factor = factor * gamma
+        out[ti] = g
+    return out.squeeze(1) if squeeze else out
+
+
+def vtrace(
+    behavior_log_probs: t.Tensor,
+    target_log_probs: t.Tensor,
+    rewards: t.Tensor,
+    values: t.Tensor,
+    bootstrap_value: t.Tensor,
+    terminals: t.Tensor,
+    gamma: float,
+    rho_clip: float = 1.0,
+    c_clip: float = 1.0,
+    pg_rho_clip: float = 1.0,
+):
+    """IMPALA V-trace targets; shapes [T, B].
+
+    Returns (vs [T,B], pg_advantages [T,B]), both detached.
+    Reference semantics: machin/frame/algorithms/impala.py:317-371.
+    """
+    T, B = rewards.shape
+    blp = behavior_log_probs.detach().float().view(T, B)
+    tlp = target_log_probs.detach().float().view(T, B)
+    rewards = rewards.detach().float().view(T, B)
+    values = values.detach().float().view(T, B)
+    bootstrap_value = bootstrap_value.detach().float().view(B)
+    nd = 1.0 - terminals.detach().float().view(T, B)
+    if rewards.is_cuda:
+        ext = _require_ext()
+        vs, pg_adv = ext.vtrace(
+            blp.contiguous(), tlp.contiguous(), rewards.contiguous(),
+            values.contiguous(), bootstrap_value.contiguous(), nd.contiguous(),
+            float(gamma), float(rho_clip), float(c_clip), float(pg_rho_clip),
+        )
+        return vs, pg_adv
+    rho = t.exp(tlp - blp)
+    clipped_rho = rho.clamp(max=rho_clip)
+    cs = rho.clamp(max=c_clip)
+    next_values = t.cat([values[1:], bootstrap_value.unsqueeze(0)], dim=0)
+    deltas = clipped_rho * (rewards + gamma * nd * next_values - values)
+    acc = t.zeros(B, dtype=rewards.dtype)
+    vs_minus_v = t.empty_like(values)
+    for i in range(T - 1, -1, -1):
+        acc = deltas[i] + gamma * nd[i] * cs[i] * acc
+        vs_minus_v[i] = acc
+    vs = vs_minus_v + values
+    vs_next = t.cat([vs[1:], bootstrap_value.unsqueeze(0)], dim=0)
+    pg_adv = rho.clamp(max=pg_rho_clip) * (rewards + gamma * nd * vs_next - values)
+    return vs, pg_adv
+
+
+# ======================================================================
+# distributional RL: categorical (C51) projection
+# ======================================================================
+def categorical_projection(
+    next_dist: t.Tensor,
+    rewards: t.Tensor,
+    terminals: t.Tensor,
+    gamma: float,
+    v_min: float,
+    v_max: float,
+) -> t.Tensor:
+    """Project r + gamma*(1-d)*z onto the fixed support; [B, A] -> [B, A].
+
+    Reference semantics: machin/frame/algorithms/rainbow.py:221-301.
+    """
+    B, A = next_dist.shape
+    next_dist = next_dist.detach().float()
+    rewards = rewards.detach().float().view(B, 1)
+    nd = 1.0 - terminals.detach().float().view(B, 1)
+    if next_dist.is_cuda:
+        ext = _require_ext()
+        return ext.categorical_projection(
+            next_dist.contiguous(), rewards.contiguous(), nd.contiguous(),
+            float(gamma), float(v_min), float(v_max),
+        )
+    delta_z = (v_max - v_min) / (A - 1)
+    z = t.linspace(v_min, v_max, A, device=next_dist.device)
+    tz = (rewards + gamma * nd * z.view(1, A)).clamp(v_min, v_max)
+    b = (tz - v_min) / delta_z
+    lo = b.floor().long()
+    hi = b.ceil().long()
+    # when b is integral, put all mass on lo
+    same = lo == hi
+    w_lo = t.where(same, t.ones_like(b), hi.float() - b)
+    w_hi = b - lo.float()
+    proj = t.zeros_like(next_dist)
+    offset = (t.arange(B, device=next_dist.device) * A).view(B, 1)
+    proj.view(-1).index_add_(0, (lo + offset).view(-1), (next_dist * w_lo).view(-1))
+    proj.view(-1).index_add_(0, (hi + offset).view(-1), (next_dist * w_hi).view(-1))
+    return proj
+
+
+__all__ = [
+    "available",
+    "polyak_update_",
+    "discounted_returns",
+    "gae",
+    "nstep_returns",
+    "vtrace",
+    "categorical_projection",
+]

@@ -1,0 +1,248 @@
+// Python bindings for the machin_amd gfx950 kernels.
+#include <torch/extension.h>
+#include <ATen/cuda/CUDAContext.h>
+#include <c10/cuda/CUDAGuard.h>
+
+#include <cstdint>
+#include <vector>
+
+using at::Tensor;
+
+// launchers defined in the .hip translation units
+void sumtree_update_launch(float*, const int64_t*, const float*, int64_t,
+                           int64_t, int, hipStream_t);
+void sumtree_build_launch(float*, int64_t, hipStream_t);
+void sumtree_sample_launch(const float*, const float*, int64_t*, int64_t,
+                           int64_t, int, int64_t, hipStream_t);
+void discounted_returns_launch(const float*, const float*, const float*,
+                               float*, int64_t, int64_t, float, hipStream_t);
+void gae_launch(const float*, const float*, const float*, const float*,
+                float*, int64_t, int64_t, float, float, hipStream_t);
+void vtrace_launch(const float*, const float*, const float*, const float*,
+                   const float*, const float*, float*, float*, int64_t,
+                   int64_t, float, float, float, float, hipStream_t);
+void categorical_projection_launch(const float*, const float*, const float*,
+                                   float*, int64_t, int64_t, float, float,
+                                   float, hipStream_t);
+void multi_tensor_polyak_launch(void*, const int64_t*, int64_t, int64_t,
+                                float, hipStream_t);
+void gaussian_sample_logprob_launch(const float*, const float*, float*,
+                                    float*, int64_t, int64_t, uint64_t,
+                                    uint64_t, int, float, hipStream_t);
+void gaussian_logprob_launch(const float*, const float*, const float*,
+                             float*, int64_t, int64_t, int, float,
+                             hipStream_t);
+void normal_noise_launch(float*, int64_t, float, float, uint64_t, uint64_t,
+                         int, hipStream_t);
+void ou_update_launch(float*, int64_t, float, float, float, float, uint64_t,
+                      uint64_t, hipStream_t);
+
+namespace {
+
+void check_f32_cuda(const Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be a CUDA tensor");
+  TORCH_CHECK(t.scalar_type() == at::kFloat, name, " must be float32");
+  TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+}
+
+hipStream_t current_stream() {
+  return at::cuda::getCurrentCUDAStream().stream();
+}
+
+// ------------------------------------------------------------------
+// sum-tree
+// ------------------------------------------------------------------
+void sumtree_update(Tensor tree, Tensor idx, Tensor w, int64_t capacity,
+                    int64_t depth) {
+  check_f32_cuda(tree, "tree");
+  check_f32_cuda(w, "w");
+  TORCH_CHECK(idx.is_cuda() && idx.scalar_type() == at::kLong &&
+                  idx.is_contiguous(),
+              "idx must be contiguous int64 CUDA");
+  const at::cuda::OptionalCUDAGuard guard(tree.device());
+  sumtree_update_launch(tree.data_ptr<float>(), idx.data_ptr<int64_t>(),
+                        w.data_ptr<float>(), idx.numel(), capacity,
+                        (int)depth, current_stream());
+}
+
+void sumtree_build(Tensor tree, int64_t capacity) {
+  check_f32_cuda(tree, "tree");
+  const at::cuda::OptionalCUDAGuard guard(tree.device());
+  sumtree_build_launch(tree.data_ptr<float>(), capacity, current_stream());
+}
+
+Tensor sumtree_sample(Tensor tree, Tensor u, int64_t capacity, int64_t depth,
+                      int64_t size) {
+  check_f32_cuda(tree, "tree");
+  check_f32_cuda(u, "u");
+  const at::cuda::OptionalCUDAGuard guard(tree.device());
+  Tensor out = at::empty({u.numel()}, u.options().dtype(at::kLong));
+  sumtree_sample_launch(tree.data_ptr<float>(), u.data_ptr<float>(),
+                        out.data_ptr<int64_t>(), u.numel(), capacity,
+                        (int)depth, size, current_stream());
+  return out;
+}
+
+// ------------------------------------------------------------------
+// scans
+// ------------------------------------------------------------------
+Tensor discounted_returns(Tensor rew, Tensor nd, Tensor bootstrap,
+                          double gamma) {
+  check_f32_cuda(rew, "rewards");
+  const at::cuda::OptionalCUDAGuard guard(rew.device());
+  int64_t T = rew.size(0), B = rew.size(1);
+  Tensor out = at::empty_like(rew);
+  discounted_returns_launch(rew.data_ptr<float>(), nd.data_ptr<float>(),
+                            bootstrap.data_ptr<float>(),
+                            out.data_ptr<float>(), T, B, (float)gamma,
+                            current_stream());
+  return out;
+}
+
+Tensor gae(Tensor rew, Tensor val, Tensor next_val, Tensor nd, double gamma,
+           double lam) {
+  check_f32_cuda(rew, "rewards");
+  const at::cuda::OptionalCUDAGuard guard(rew.device());
+  int64_t T = rew.size(0), B = rew.size(1);
+  Tensor out = at::empty_like(rew);
+  gae_launch(rew.data_ptr<float>(), val.data_ptr<float>(),
+             next_val.data_ptr<float>(), nd.data_ptr<float>(),
+             out.data_ptr<float>(), T, B, (float)gamma, (float)lam,
+             current_stream());
+  return out;
+}
+
+std::vector<Tensor> vtrace(Tensor blp, Tensor tlp, Tensor rew, Tensor val,
+                           Tensor bootstrap, Tensor nd, double gamma,
+                           double rho_clip, double c_clip,
+                           double pg_rho_clip) {
+  check_f32_cuda(rew, "rewards");
+  const at::cuda::OptionalCUDAGuard guard(rew.device());
+  int64_t T = rew.size(0), B = rew.size(1);
+  Tensor vs = at::empty_like(rew);
+  Tensor pg_adv = at::empty_like(rew);
+  vtrace_launch(blp.data_ptr<float>(), tlp.data_ptr<float>(),
+                rew.data_ptr<float>(), val.data_ptr<float>(),
+                bootstrap.data_ptr<float>(), nd.data_ptr<float>(),
+                vs.data_ptr<float>(), pg_adv.data_ptr<float>(), T, B,
+                (float)gamma, (float)rho_clip, (float)c_clip,
+                (float)pg_rho_clip, current_stream());
+  return {vs, pg_adv};
+}
+
+// ------------------------------------------------------------------
+// categorical projection
+// ------------------------------------------------------------------
+Tensor categorical_projection(Tensor next_dist, Tensor rew, Tensor nd,
+                              double gamma, double v_min, double v_max) {
+  check_f32_cuda(next_dist, "next_dist");
+  const at::cuda::OptionalCUDAGuard guard(next_dist.device());
+  int64_t B = next_dist.size(0), A = next_dist.size(1);
+  TORCH_CHECK(A <= 256, "categorical projection supports at most 256 atoms");
+  Tensor out = at::empty_like(next_dist);
+  categorical_projection_launch(next_dist.data_ptr<float>(),
+                                rew.data_ptr<float>(), nd.data_ptr<float>(),
+                                out.data_ptr<float>(), B, A, (float)gamma,
+                                (float)v_min, (float)v_max,
+                                current_stream());
+  return out;
+}
+
+// ------------------------------------------------------------------
+// multi-tensor polyak
+// ------------------------------------------------------------------
+void multi_tensor_polyak(std::vector<Tensor> targets,
+                         std::vector<Tensor> sources, double tau) {
+  TORCH_CHECK(targets.size() == sources.size(), "list size mismatch");
+  if (targets.empty()) return;
+  const at::cuda::OptionalCUDAGuard guard(targets[0].device());
+  int64_t n = (int64_t)targets.size();
+  // host-side table: [tgt_ptr, src_ptr] pairs then exclusive prefix
+  Tensor table = at::empty({n * 2 + n + 1},
+                           at::TensorOptions().dtype(at::kLong));
+  int64_t* h = table.data_ptr<int64_t>();
+  int64_t total = 0;
+  for (int64_t i = 0; i < n; ++i) {
+    check_f32_cuda(targets[i], "target");
+    check_f32_cuda(sources[i], "source");
+    TORCH_CHECK(targets[i].numel() == sources[i].numel(), "numel mismatch");
+    h[2 * i] = (int64_t)targets[i].data_ptr<float>();
+    h[2 * i + 1] = (int64_t)sources[i].data_ptr<float>();
+    h[2 * n + i] = total;
+    total += targets[i].numel();
+  }
+  h[2 * n + n] = total;
+  Tensor dev_table = table.to(targets[0].device(), /*non_blocking=*/true);
+  int64_t* d = dev_table.data_ptr<int64_t>();
+  multi_tensor_polyak_launch((void*)d, d + 2 * n, n, total, (float)tau,
+                             current_stream());
+}
+
+// ------------------------------------------------------------------
+// distributions
+// ------------------------------------------------------------------
+std::vector<Tensor> gaussian_sample_logprob(Tensor mu, Tensor log_std,
+                                            int64_t seed, int64_t offset,
+                                            bool tanh_squash,
+                                            double epsilon) {
+  check_f32_cuda(mu, "mu");
+  check_f32_cuda(log_std, "log_std");
+  const at::cuda::OptionalCUDAGuard guard(mu.device());
+  int64_t B = mu.size(0), D = mu.size(1);
+  Tensor act = at::empty_like(mu);
+  Tensor logp = at::empty({B, 1}, mu.options());
+  gaussian_sample_logprob_launch(
+      mu.data_ptr<float>(), log_std.data_ptr<float>(), act.data_ptr<float>(),
+      logp.data_ptr<float>(), B, D, (uint64_t)seed, (uint64_t)offset,
+      tanh_squash ? 1 : 0, (float)epsilon, current_stream());
+  return {act, logp};
+}
+
+Tensor gaussian_logprob(Tensor mu, Tensor log_std, Tensor act,
+                        bool tanh_squash, double epsilon) {
+  check_f32_cuda(mu, "mu");
+  const at::cuda::OptionalCUDAGuard guard(mu.device());
+  int64_t B = mu.size(0), D = mu.size(1);
+  Tensor logp = at::empty({B, 1}, mu.options());
+  gaussian_logprob_launch(mu.data_ptr<float>(), log_std.data_ptr<float>(),
+                          act.data_ptr<float>(), logp.data_ptr<float>(), B,
+                          D, tanh_squash ? 1 : 0, (float)epsilon,
+                          current_stream());
+  return logp;
+}
+
+void normal_noise_(Tensor x, double mean, double std, int64_t seed,
+                   int64_t offset, bool add) {
+  check_f32_cuda(x, "x");
+  const at::cuda::OptionalCUDAGuard guard(x.device());
+  normal_noise_launch(x.data_ptr<float>(), x.numel(), (float)mean,
+                      (float)std, (uint64_t)seed, (uint64_t)offset,
+                      add ? 1 : 0, current_stream());
+}
+
+void ou_update_(Tensor x, double mu, double theta, double sigma, double dt,
+                int64_t seed, int64_t offset) {
+  check_f32_cuda(x, "x");
+  const at::cuda::OptionalCUDAGuard guard(x.device());
+  ou_update_launch(x.data_ptr<float>(), x.numel(), (float)mu, (float)theta,
+                   (float)sigma, (float)dt, (uint64_t)seed,
+                   (uint64_t)offset, current_stream());
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.doc() = "machin_amd gfx950 HIP kernels";
+  m.def("sumtree_update", &sumtree_update);
+  m.def("sumtree_build", &sumtree_build);
+  m.def("sumtree_sample", &sumtree_sample);
+  m.def("discounted_returns", &discounted_returns);
+  m.def("gae", &gae);
+  m.def("vtrace", &vtrace);
+  m.def("categorical_projection", &categorical_projection);
+  m.def("multi_tensor_polyak", &multi_tensor_polyak);
+  m.def("gaussian_sample_logprob", &gaussian_sample_logprob);
+  m.def("gaussian_logprob", &gaussian_logprob);
+  m.def("normal_noise_", &normal_noise_);
+  m.def("ou_update_", &ou_update_);
+}

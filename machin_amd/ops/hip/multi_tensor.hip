@@ -1,0 +1,62 @@
+// gfx950 fused multi-tensor ops: polyak soft-update (and general
+// axpby) over a whole parameter list in ONE kernel launch.
+//
+// Replaces the per-parameter python loop of the reference
+// (machin/frame/algorithms/utils.py:8-27). HBM-bound: 2 reads + 1
+// write per element, one pass over all tensors.
+//
+// The tensor table (pointer pairs + exclusive prefix of element
+// counts) lives in a device buffer; each block binary-searches the
+// chunk it owns (guide §6 G11: grid-stride, blocks capped).
+#include "common.h"
+
+struct MTEntry {
+  float* tgt;
+  const float* src;
+};
+
+// chunk c covers elements [c*CHUNK, ...) of the virtual concatenation.
+#define MT_CHUNK 65536
+#define MT_BLOCK 256
+
+template <typename T>
+__device__ __forceinline__ float ma_to_float(T v);
+template <>
+__device__ __forceinline__ float ma_to_float<float>(float v) { return v; }
+
+__global__ void multi_tensor_polyak_kernel(
+    MTEntry* __restrict__ entries, const int64_t* __restrict__ prefix,
+    int64_t n_tensors, int64_t total, float tau) {
+  const float keep = 1.0f - tau;
+  for (int64_t start = (int64_t)blockIdx.x * MT_CHUNK; start < total;
+       start += (int64_t)gridDim.x * MT_CHUNK) {
+    // binary search the tensor containing `start`
+    int64_t lo = 0, hi = n_tensors - 1;
+    while (lo < hi) {
+      int64_t mid = (lo + hi + 1) >> 1;
+      if (prefix[mid] <= start) lo = mid; else hi = mid - 1;
+    }
+    int64_t ti = lo;
+    int64_t chunk_end = min(start + (int64_t)MT_CHUNK, total);
+    int64_t i = start + threadIdx.x;
+    while (i < chunk_end) {
+      // advance tensor index when crossing a boundary
+      while (ti + 1 < n_tensors && prefix[ti + 1] <= i) ++ti;
+      int64_t off = i - prefix[ti];
+      float* tgt = entries[ti].tgt;
+      const float* src = entries[ti].src;
+      tgt[off] = keep * tgt[off] + tau * src[off];
+      i += MT_BLOCK;
+    }
+  }
+}
+
+void multi_tensor_polyak_launch(void* entries, const int64_t* prefix,
+                                int64_t n_tensors, int64_t total, float tau,
+                                hipStream_t stream) {
+  if (total == 0) return;
+  int grid = ma_grid((total + MT_CHUNK - 1) / MT_CHUNK, 1);
+  hipLaunchKernelGGL(multi_tensor_polyak_kernel, dim3(grid), dim3(MT_BLOCK),
+                     0, stream, (MTEntry*)entries, prefix, n_tensors, total,
+                     tau);
+}

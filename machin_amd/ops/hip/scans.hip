@@ -1,0 +1,110 @@
+// gfx950 reverse-time scan kernels: discounted returns, GAE, V-trace.
+//
+// Reference semantics being replaced (python loops):
+//   machin/frame/algorithms/a2c.py:269-326   (GAE / discounted return)
+//   machin/frame/algorithms/impala.py:317-371 (V-trace)
+//
+// Layout: [T, B] row-major, so at fixed t consecutive b are contiguous
+// -> each thread owns one batch column b and the per-step loads of a
+// wave are fully coalesced (guide §2). T is small (rollout length),
+// B*num_cols fills the chip; per-thread sequential recursion over T is
+// the natural CDNA formulation (no cross-lane dependency at all).
+#include "common.h"
+
+__global__ void discounted_returns_kernel(const float* __restrict__ rew,
+                                          const float* __restrict__ nd,
+                                          const float* __restrict__ bootstrap,
+                                          float* __restrict__ out, int64_t T,
+                                          int64_t B, float gamma) {
+  for (int64_t b = blockIdx.x * blockDim.x + threadIdx.x; b < B;
+       b += (int64_t)gridDim.x * blockDim.x) {
+    float r = bootstrap[b];
+    for (int64_t t = T - 1; t >= 0; --t) {
+      int64_t k = t * B + b;
+      r = rew[k] + gamma * nd[k] * r;
+      out[k] = r;
+    }
+  }
+}
+
+__global__ void gae_kernel(const float* __restrict__ rew,
+                           const float* __restrict__ val,
+                           const float* __restrict__ next_val,
+                           const float* __restrict__ nd,
+                           float* __restrict__ out, int64_t T, int64_t B,
+                           float gamma, float lam) {
+  for (int64_t b = blockIdx.x * blockDim.x + threadIdx.x; b < B;
+       b += (int64_t)gridDim.x * blockDim.x) {
+    float acc = 0.0f;
+    for (int64_t t = T - 1; t >= 0; --t) {
+      int64_t k = t * B + b;
+      float ndk = nd[k];
+      float delta = rew[k] + gamma * ndk * next_val[k] - val[k];
+      acc = delta + gamma * lam * ndk * acc;
+      out[k] = acc;
+    }
+  }
+}
+
+// V-trace: one backward pass producing both vs and pg advantages.
+// carry vs[t+1] while walking t = T-1 .. 0.
+__global__ void vtrace_kernel(const float* __restrict__ blp,
+                              const float* __restrict__ tlp,
+                              const float* __restrict__ rew,
+                              const float* __restrict__ val,
+                              const float* __restrict__ bootstrap,
+                              const float* __restrict__ nd,
+                              float* __restrict__ vs_out,
+                              float* __restrict__ pg_adv, int64_t T,
+                              int64_t B, float gamma, float rho_clip,
+                              float c_clip, float pg_rho_clip) {
+  for (int64_t b = blockIdx.x * blockDim.x + threadIdx.x; b < B;
+       b += (int64_t)gridDim.x * blockDim.x) {
+    float acc = 0.0f;             // vs[t] - V[t] carry
+    float vs_next = bootstrap[b]; // vs[t+1]
+    float v_next = bootstrap[b];  // V[t+1]
+    for (int64_t t = T - 1; t >= 0; --t) {
+      int64_t k = t * B + b;
+      float rho = __expf(tlp[k] - blp[k]);
+      float ndk = nd[k];
+      float vk = val[k];
+      float td = rew[k] + gamma * ndk * v_next - vk;
+      float delta = fminf(rho, rho_clip) * td;
+      acc = delta + gamma * ndk * fminf(rho, c_clip) * acc;
+      float vs_t = acc + vk;
+      pg_adv[k] =
+          fminf(rho, pg_rho_clip) * (rew[k] + gamma * ndk * vs_next - vk);
+      vs_out[k] = vs_t;
+      vs_next = vs_t;
+      v_next = vk;
+    }
+  }
+}
+
+void discounted_returns_launch(const float* rew, const float* nd,
+                               const float* bootstrap, float* out, int64_t T,
+                               int64_t B, float gamma, hipStream_t stream) {
+  const int block = 256;
+  hipLaunchKernelGGL(discounted_returns_kernel, dim3(ma_grid(B, block)),
+                     dim3(block), 0, stream, rew, nd, bootstrap, out, T, B,
+                     gamma);
+}
+
+void gae_launch(const float* rew, const float* val, const float* next_val,
+                const float* nd, float* out, int64_t T, int64_t B, float gamma,
+                float lam, hipStream_t stream) {
+  const int block = 256;
+  hipLaunchKernelGGL(gae_kernel, dim3(ma_grid(B, block)), dim3(block), 0,
+                     stream, rew, val, next_val, nd, out, T, B, gamma, lam);
+}
+
+void vtrace_launch(const float* blp, const float* tlp, const float* rew,
+                   const float* val, const float* bootstrap, const float* nd,
+                   float* vs_out, float* pg_adv, int64_t T, int64_t B,
+                   float gamma, float rho_clip, float c_clip,
+                   float pg_rho_clip, hipStream_t stream) {
+  const int block = 256;
+  hipLaunchKernelGGL(vtrace_kernel, dim3(ma_grid(B, block)), dim3(block), 0,
+                     stream, blp, tlp, rew, val, bootstrap, nd, vs_out,
+                     pg_adv, T, B, gamma, rho_clip, c_clip, pg_rho_clip);
+}

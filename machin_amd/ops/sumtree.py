@@ -1,0 +1,88 @@
+"""Device-resident sum-tree (GPU twin of frame.buffers.WeightTree).
+
+Same heap layout as the CPU tree: one float32 tensor of length
+2*capacity, root at index 1, leaves at [capacity, 2*capacity). All
+operations are HIP kernels (machin_amd/ops/hip/sumtree.hip); nothing
+synchronizes with the host except explicit scalar reads.
+
+Reference CPU baselines this must beat (BASELINE.md / reference
+machin/frame/buffers/prioritized_buffer.py:32-40): build 10M leaves
+90 ms, 10M lookups 230 ms, 1M batched update 20 ms.
+"""
+import math
+from typing import Union
+
+import torch as t
+
+from . import _require_ext
+
+
+class DeviceSumTree:
+    """Sum-tree over ``size`` leaves stored on a ROCm device."""
+
+    def __init__(self, size: int, device: Union[str, t.device] = "cuda:0"):
+        if size <= 0:
+            raise ValueError("Tree size must be positive.")
+        self.size = int(size)
+        self.depth = max(1, int(math.ceil(math.log2(max(self.size, 2)))))
+        self.capacity = 1 << self.depth
+        self.device = t.device(device)
+        self.weights = t.zeros(
+            2 * self.capacity, dtype=t.float32, device=self.device
+        )
+        self._ext = _require_ext()
+
+    # -- accessors -----------------------------------------------------
+    def get_weight_sum(self) -> float:
+        return float(self.weights[1].item())
+
+    def get_weight_sum_tensor(self) -> t.Tensor:
+        """Root weight as a 0-dim device tensor (no host sync)."""
+        return self.weights[1]
+
+    def get_leaf_all_weights(self) -> t.Tensor:
+        return self.weights[self.capacity : self.capacity + self.size]
+
+    def get_leaf_weight(self, index: t.Tensor) -> t.Tensor:
+        index = index.to(device=self.device, dtype=t.long)
+        return self.weights[self.capacity + index]
+
+    def get_leaf_max(self) -> float:
+        if self.size == 0:
+            return 0.0
+        return float(self.get_leaf_all_weights().max().item())
+
+    # -- updates -------------------------------------------------------
+    def update_leaf_batch(self, weights: t.Tensor, indexes: t.Tensor):
+        weights = weights.to(device=self.device, dtype=t.float32).contiguous()
+        indexes = indexes.to(device=self.device, dtype=t.long).contiguous()
+        self._ext.sumtree_update(
+            self.weights, indexes, weights, self.capacity, self.depth
+        )
+
+    def update_all_leaves(self, weights: t.Tensor):
+        if weights.numel() != self.size:
+            raise ValueError(f"Expected {self.size} weights.")
+        leaves = self.weights[self.capacity : self.capacity + self.size]
+        leaves.copy_(weights.to(device=self.device, dtype=t.float32))
+        self._ext.sumtree_build(self.weights, self.capacity)
+
+    # -- queries -------------------------------------------------------
+    def find_leaf_index(self, prefix_weights: t.Tensor) -> t.Tensor:
+        """Map prefix weights in [0, sum) to leaf indexes."""
+        u = prefix_weights.to(device=self.device, dtype=t.float32).contiguous()
+        return self._ext.sumtree_sample(
+            self.weights, u, self.capacity, self.depth, self.size
+        )
+
+    def sample(self, n: int, stratified: bool = True) -> t.Tensor:
+        """Draw ``n`` leaves proportional to weight, fully on device."""
+        total = self.get_weight_sum_tensor()
+        if stratified:
+            u = (
+                t.rand(n, device=self.device, dtype=t.float32)
+                + t.arange(n, device=self.device, dtype=t.float32)
+            ) * (total / n)
+        else:
+            u = t.rand(n, device=self.device, dtype=t.float32) * total
+        return self.find_leaf_index(u)

@@ -1,0 +1,214 @@
+"""GPU numerics: every gfx950 HIP kernel vs a plain fp32 PyTorch
+reference of the same op. All tests gpu-marked."""
+import numpy as np
+import pytest
+import torch as t
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.fixture(scope="module")
+def dev():
+    if not t.cuda.is_available():
+        pytest.skip("no GPU")
+    import machin_amd.ops as ops
+
+    assert ops.available(), "HIP extension must be built on a GPU box"
+    return t.device("cuda:0")
+
+
+class TestSumTreeGPU:
+    def test_build_and_sum(self, dev):
+        from machin_amd.ops.sumtree import DeviceSumTree
+
+        for size in (1, 5, 1000, 1 << 17):
+            tree = DeviceSumTree(size, dev)
+            w = t.rand(size)
+            tree.update_all_leaves(w)
+            assert tree.get_weight_sum() == pytest.approx(
+                float(w.sum()), rel=1e-4
+            )
+
+    def test_update_batch(self, dev):
+        from machin_amd.ops.sumtree import DeviceSumTree
+
+        size = 10000
+        tree = DeviceSumTree(size, dev)
+        w = t.rand(size)
+        tree.update_all_leaves(w)
+        idx = t.randint(0, size, (500,))
+        idx = t.unique(idx)
+        new_w = t.rand(idx.numel()) + 1.0
+        tree.update_leaf_batch(new_w, idx)
+        expect = w.clone()
+        expect[idx] = new_w
+        assert tree.get_weight_sum() == pytest.approx(
+            float(expect.sum()), rel=1e-4
+        )
+        assert t.allclose(
+            tree.get_leaf_weight(idx).cpu(), new_w, atol=1e-6
+        )
+
+    def test_find_leaf_matches_cpu(self, dev):
+        from machin_amd.frame.buffers import WeightTree
+        from machin_amd.ops.sumtree import DeviceSumTree
+
+        size = 4096
+        w = np.random.uniform(0.1, 2.0, size)
+        cpu = WeightTree(size)
+        cpu.update_all_leaves(w)
+        gpu = DeviceSumTree(size, dev)
+        gpu.update_all_leaves(t.tensor(w, dtype=t.float32))
+        queries = np.random.uniform(0, cpu.get_weight_sum() * 0.999, 1000)
+        cpu_idx = cpu.find_leaf_index(queries)
+        gpu_idx = gpu.find_leaf_index(t.tensor(queries, dtype=t.float32))
+        # fp32 rounding can shift a query across one leaf boundary:
+        # allow off-by-one leaves but nothing else
+        diff = np.abs(gpu_idx.cpu().numpy() - cpu_idx)
+        assert (diff <= 1).all()
+        assert (diff == 0).mean() > 0.98
+
+    def test_sample_distribution(self, dev):
+        from machin_amd.ops.sumtree import DeviceSumTree
+
+        size = 64
+        tree = DeviceSumTree(size, dev)
+        w = t.rand(size) + 0.05
+        tree.update_all_leaves(w)
+        idx = tree.sample(20000, stratified=False)
+        counts = t.bincount(idx.cpu(), minlength=size).float() / 20000
+        assert (counts - w / w.sum()).abs().max() < 0.02
+
+
+class TestScansGPU:
+    def test_discounted_returns(self, dev):
+        import machin_amd.ops as ops
+
+        T, B = 64, 128
+        rew = t.rand(T, B)
+        term = (t.rand(T, B) > 0.9).float()
+        boot = t.rand(B)
+        cpu = ops.discounted_returns(rew, term, 0.97, boot)
+        gpu = ops.discounted_returns(
+            rew.to(dev), term.to(dev), 0.97, boot.to(dev)
+        )
+        assert t.allclose(gpu.cpu(), cpu, atol=1e-5)
+
+    def test_gae(self, dev):
+        import machin_amd.ops as ops
+
+        T, B = 40, 96
+        rew, val, nxt = t.rand(T, B), t.rand(T, B), t.rand(T, B)
+        term = (t.rand(T, B) > 0.85).float()
+        cpu = ops.gae(rew, val, nxt, term, 0.99, 0.95)
+        gpu = ops.gae(rew.to(dev), val.to(dev), nxt.to(dev), term.to(dev),
+                      0.99, 0.95)
+        assert t.allclose(gpu.cpu(), cpu, atol=1e-5)
+
+    def test_vtrace(self, dev):
+        import machin_amd.ops as ops
+
+        T, B = 32, 64
+        blp = -t.rand(T, B)
+        tlp = -t.rand(T, B)
+        rew, val = t.rand(T, B), t.rand(T, B)
+        boot = t.rand(B)
+        term = (t.rand(T, B) > 0.9).float()
+        cpu_vs, cpu_pg = ops.vtrace(blp, tlp, rew, val, boot, term, 0.99)
+        gpu_vs, gpu_pg = ops.vtrace(
+            blp.to(dev), tlp.to(dev), rew.to(dev), val.to(dev), boot.to(dev),
+            term.to(dev), 0.99,
+        )
+        assert t.allclose(gpu_vs.cpu(), cpu_vs, atol=1e-4)
+        assert t.allclose(gpu_pg.cpu(), cpu_pg, atol=1e-4)
+
+
+class TestProjectionGPU:
+    def test_vs_cpu(self, dev):
+        import machin_amd.ops as ops
+
+        B, A = 256, 51
+        dist = t.softmax(t.randn(B, A), dim=1)
+        rew = t.randn(B) * 4
+        term = (t.rand(B) > 0.8).float()
+        cpu = ops.categorical_projection(dist, rew, term, 0.99, -10, 10)
+        gpu = ops.categorical_projection(
+            dist.to(dev), rew.to(dev), term.to(dev), 0.99, -10, 10
+        )
+        assert t.allclose(gpu.cpu(), cpu, atol=1e-5)
+        assert t.allclose(gpu.sum(dim=1).cpu(), t.ones(B), atol=1e-5)
+
+
+class TestPolyakGPU:
+    def test_vs_cpu(self, dev):
+        import machin_amd.ops as ops
+
+        shapes = [(128, 64), (1000,), (3, 3, 16), (1,)]
+        tgt = [t.rand(*s, device=dev) for s in shapes]
+        src = [t.rand(*s, device=dev) for s in shapes]
+        expect = [a * 0.995 + b * 0.005 for a, b in zip(tgt, src)]
+        ops.polyak_update_(tgt, src, 0.005)
+        for a, e in zip(tgt, expect):
+            assert t.allclose(a, e, atol=1e-6)
+
+
+class TestDistributionsGPU:
+    def test_gaussian_sample_logprob(self, dev):
+        from machin_amd.ops import _require_ext
+
+        ext = _require_ext()
+        B, D = 20000, 4
+        mu = t.randn(B, D, device=dev)
+        log_std = t.full((B, D), -0.5, device=dev)
+        act, logp = ext.gaussian_sample_logprob(mu, log_std, 1234, 0, False,
+                                                1e-6)
+        # statistics of samples
+        err = (act - mu).mean().abs().item()
+        assert err < 0.02
+        std_err = abs((act - mu).std().item() - float(t.exp(t.tensor(-0.5))))
+        assert std_err < 0.02
+        # log-prob must match torch.distributions exactly given the action
+        d = t.distributions.Normal(mu, log_std.exp())
+        expect = d.log_prob(act).sum(dim=1, keepdim=True)
+        assert t.allclose(logp, expect, atol=1e-3)
+
+    def test_tanh_gaussian(self, dev):
+        from machin_amd.ops import _require_ext
+
+        ext = _require_ext()
+        B, D = 4096, 6
+        mu = t.randn(B, D, device=dev) * 0.3
+        log_std = t.full((B, D), -1.0, device=dev)
+        act, logp = ext.gaussian_sample_logprob(mu, log_std, 99, 7, True,
+                                                1e-6)
+        assert act.abs().max().item() <= 1.0
+        u = t.atanh(act.clamp(-1 + 1e-6, 1 - 1e-6))
+        d = t.distributions.Normal(mu, log_std.exp())
+        expect = (
+            d.log_prob(u) - t.log(1 - act * act + 1e-6)
+        ).sum(dim=1, keepdim=True)
+        assert t.allclose(logp, expect, atol=5e-3, rtol=1e-3)
+
+    def test_gaussian_logprob_of_given(self, dev):
+        from machin_amd.ops import _require_ext
+
+        ext = _require_ext()
+        B, D = 512, 3
+        mu = t.randn(B, D, device=dev)
+        log_std = t.randn(B, D, device=dev) * 0.2
+        act = t.randn(B, D, device=dev)
+        logp = ext.gaussian_logprob(mu, log_std, act, False, 1e-6)
+        d = t.distributions.Normal(mu, log_std.exp())
+        expect = d.log_prob(act).sum(dim=1, keepdim=True)
+        assert t.allclose(logp, expect, atol=1e-3)
+
+    def test_ou_mean_reversion(self, dev):
+        from machin_amd.ops import _require_ext
+
+        ext = _require_ext()
+        x = t.full((100000,), 5.0, device=dev)
+        for step in range(200):
+            ext.ou_update_(x, 0.0, 0.15, 0.2, 1.0, 42, step)
+        # stationary: mean ~ 0, std ~ sigma/sqrt(2*theta) = 0.365
+        assert abs(x.mean().item()) < 0.05
+        assert abs(x.std().item() - 0.2 / (2 * 0.15) ** 0.5) < 0.05

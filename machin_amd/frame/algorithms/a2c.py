@@ -1,0 +1,352 @@
+"""A2C: advantage actor-critic.
+
+Parity target: reference ``machin/frame/algorithms/a2c.py`` (:57-422):
+actor contract ``forward(state[, action]) -> (action, log_prob[,
+entropy])``; GAE / discounted-return computed at ``store_episode`` time
+(:269-326) — here through the gfx950 scan kernels
+(machin_amd.ops.gae / discounted_returns) when the critic lives on
+GPU; policy loss ``-log_prob * advantage``, entropy bonus, weighted
+value loss; buffer cleared after each update.
+"""
+from typing import Any, Callable, Dict, List, Union
+
+import numpy as np
+import torch as t
+import torch.nn as nn
+
+from ... import ops
+from ...utils.conf import Config
+from ..buffers.buffer import Buffer
+from ..transition import Transition
+from .base import TorchFramework
+from .utils import (
+    assert_and_get_valid_criterion,
+    assert_and_get_valid_lr_scheduler,
+    assert_and_get_valid_models,
+    assert_and_get_valid_optimizer,
+    safe_call,
+    safe_return,
+)
+
+
+class A2C(TorchFramework):
+    _is_top = ["actor", "critic"]
+    _is_restorable = ["actor", "critic"]
+
+    def __init__(
+        self,
+        actor: nn.Module,
+        critic: nn.Module,
+        optimizer: Callable,
+        criterion: Callable,
+        *_,
+        lr_scheduler: Callable = None,
+        lr_scheduler_args: tuple = None,
+        lr_scheduler_kwargs: tuple = None,
+        batch_size: int = 100,
+        actor_update_times: int = 5,
+        critic_update_times: int = 10,
+        actor_learning_rate: float = 0.001,
+        critic_learning_rate: float = 0.001,
+        entropy_weight: float = None,
+        value_weight: float = 0.5,
+        gradient_max: float = np.inf,
+        gae_lambda: float = 1.0,
+        discount: float = 0.99,
+        normalize_advantage: bool = True,
+        replay_size: int = 500000,
+        replay_device: Union[str, t.device] = "cpu",
+        replay_buffer: Buffer = None,
+        visualize: bool = False,
+        visualize_dir: str = "",
+        **__,
+    ):
+        super().__init__()
+        self.batch_size = batch_size
+        self.actor_update_times = actor_update_times
+        self.critic_update_times = critic_update_times
+        self.entropy_weight = entropy_weight
+        self.value_weight = value_weight
+        self.grad_max = gradient_max
+        self.gae_lambda = gae_lambda
+        self.discount = discount
+        self.normalize_advantage = normalize_advantage
+        self.visualize = visualize
+        self.visualize_dir = visualize_dir
+
+        self.actor = actor
+        self.critic = critic
+        if isinstance(optimizer, (tuple, list)):
+            a_opt, c_opt = optimizer
+        else:
+            a_opt = c_opt = optimizer
+        self.actor_optim = a_opt(self.actor.parameters(), lr=actor_learning_rate)
+        self.critic_optim = c_opt(
+            self.critic.parameters(), lr=critic_learning_rate
+        )
+        self.replay_buffer = (
+            Buffer(replay_size, replay_device)
+            if replay_buffer is None
+            else replay_buffer
+        )
+        self.actor_lr_sch = None
+        self.critic_lr_sch = None
+        if lr_scheduler is not None:
+            args = lr_scheduler_args or ((), ())
+            kwargs = lr_scheduler_kwargs or ({}, {})
+            self.actor_lr_sch = lr_scheduler(self.actor_optim, *args[0], **kwargs[0])
+            self.critic_lr_sch = lr_scheduler(self.critic_optim, *args[1], **kwargs[1])
+        self.criterion = criterion() if isinstance(criterion, type) else criterion
+
+    @property
+    def optimizers(self):
+        return [self.actor_optim, self.critic_optim]
+
+    @optimizers.setter
+    def optimizers(self, optimizers):
+        self.actor_optim, self.critic_optim = optimizers
+
+    @property
+    def lr_schedulers(self):
+        out = []
+        if self.actor_lr_sch is not None:
+            out.append(self.actor_lr_sch)
+        if self.critic_lr_sch is not None:
+            out.append(self.critic_lr_sch)
+        return out
+
+    # ------------------------------------------------------------------
+    def act(self, state: Dict[str, Any], *_, **__):
+        """Sample an action: returns the actor's (action, log_prob[,
+        entropy]) tuple."""
+        result = safe_call(self.actor, state)
+        if not isinstance(result, tuple):
+            raise ValueError(
+                "A2C actor must return a tuple (action, log_prob[, entropy])."
+            )
+        return result
+
+    def _eval_act(self, state: Dict[str, Any], action: Dict[str, Any], **__):
+        """Evaluate log-prob (and entropy) of a stored action."""
+        return safe_call(self.actor, state, action)
+
+    def _criticize(self, state: Dict[str, Any], *_, **__):
+        return safe_return(safe_call(self.critic, state))
+
+    # ------------------------------------------------------------------
+    def store_transition(self, transition: Union[Transition, Dict]):
+        """Store one transition; GAE cannot be computed until the full
+        episode arrives, so prefer store_episode."""
+        raise NotImplementedError(
+            "A2C requires whole episodes: use store_episode()."
+        )
+
+    def store_episode(self, episode: List[Union[Transition, Dict]]):
+        """Compute per-step values, GAE advantages and discounted
+        targets for the episode, then store."""
+        episode = [
+            Transition(**tr) if isinstance(tr, dict) else tr for tr in episode
+        ]
+        T = len(episode)
+        with t.no_grad():
+            states = {}
+            for key in episode[0].state.keys():
+                states[key] = t.cat([tr.state[key] for tr in episode], dim=0)
+            values = self._criticize(states).view(T)
+            device = values.device
+            last_next = {
+                k: v for k, v in episode[-1].next_state.items()
+            }
+            rewards = t.tensor(
+                [float(tr.reward) if not t.is_tensor(tr.reward)
+                 else float(tr.reward.reshape(-1)[0]) for tr in episode],
+                device=device,
+            )
+            terminals = t.tensor(
+                [float(tr.terminal) if not t.is_tensor(tr.terminal)
+                 else float(tr.terminal.reshape(-1)[0]) for tr in episode],
+                device=device,
+            )
+            # bootstrap from critic when the episode was truncated
+            if terminals[-1] < 0.5:
+                boot = self._criticize(last_next).view(-1)[0]
+            else:
+                boot = t.zeros((), device=device)
+            next_values = t.cat([values[1:], boot.view(1)])
+            # correct bootstrap inside the episode too: next value at a
+            # terminal step is masked by (1-terminal) inside the scans
+            advantages = ops.gae(
+                rewards, values, next_values, terminals, self.discount,
+                self.gae_lambda,
+            )
+            targets = advantages + values
+
+        for i, tr in enumerate(episode):
+            tr_dict = {k: getattr(tr, k) for k in tr.keys()}
+            tr_dict["value"] = float(targets[i].item())
+            tr_dict["gae"] = float(advantages[i].item())
+            episode[i] = Transition(**tr_dict)
+        self.replay_buffer.store_episode(
+            episode,
+            required_attrs=(
+                "state", "action", "next_state", "reward", "terminal",
+                "value", "gae",
+            ),
+        )
+
+    # ------------------------------------------------------------------
+    def update(self, update_value=True, update_policy=True,
+               concatenate_samples=True, **__):
+        """Update on everything stored, then clear the buffer."""
+        (
+            batch_size,
+            (state, action, reward, next_state, terminal, target_value,
+             advantage),
+        ) = self.replay_buffer.sample_batch(
+            -1,
+            sample_method="all",
+            concatenate=concatenate_samples,
+            sample_attrs=[
+                "state", "action", "reward", "next_state", "terminal",
+                "value", "gae",
+            ],
+            additional_concat_custom_attrs=["value", "gae"],
+        )
+        if batch_size == 0:
+            return 0.0, 0.0
+        self.actor.train()
+        self.critic.train()
+
+        sum_act_loss = 0.0
+        sum_value_loss = 0.0
+
+        for _ in range(self.actor_update_times):
+            result = self._eval_act(state, action)
+            new_log_prob = result[1].view(batch_size, 1)
+            entropy = (
+                result[2].view(batch_size, 1) if len(result) > 2 else None
+            )
+            adv = advantage.to(new_log_prob.device).view(batch_size, 1).detach()
+            if self.normalize_advantage:
+                adv = (adv - adv.mean()) / (adv.std() + 1e-6)
+            act_policy_loss = -(new_log_prob * adv)
+            if self.entropy_weight is not None and entropy is not None:
+                act_policy_loss += self.entropy_weight * -entropy
+            act_policy_loss = act_policy_loss.mean()
+            if self.visualize:
+                self.visualize_model(
+                    act_policy_loss, "actor", self.visualize_dir
+                )
+            if update_policy:
+                self.actor_optim.zero_grad(set_to_none=True)
+                self._backward(act_policy_loss)
+                nn.utils.clip_grad_norm_(
+                    self.actor.parameters(), self.grad_max
+                )
+                self.actor_optim.step()
+            sum_act_loss += float(act_policy_loss.detach().item())
+
+        for _ in range(self.critic_update_times):
+            value = self._criticize(state).view(batch_size, 1)
+            tv = target_value.to(value.device).view(batch_size, 1)
+            value_loss = (
+                self.criterion(value, tv.to(value.dtype)) * self.value_weight
+            )
+            if self.visualize:
+                self.visualize_model(
+                    value_loss, "critic", self.visualize_dir
+                )
+            if update_value:
+                self.critic_optim.zero_grad(set_to_none=True)
+                self._backward(value_loss)
+                nn.utils.clip_grad_norm_(
+                    self.critic.parameters(), self.grad_max
+                )
+                self.critic_optim.step()
+            sum_value_loss += float(value_loss.detach().item())
+
+        self.replay_buffer.clear()
+        return (
+            -sum_act_loss / max(self.actor_update_times, 1),
+            sum_value_loss / max(self.critic_update_times, 1),
+        )
+
+    def update_lr_scheduler(self):
+        if self.actor_lr_sch is not None:
+            self.actor_lr_sch.step()
+        if self.critic_lr_sch is not None:
+            self.critic_lr_sch.step()
+
+    # ------------------------------------------------------------------
+    @classmethod
+    def generate_config(cls, config: Union[Dict[str, Any], Config]):
+        default = {
+            "frame": cls.__name__,
+            "models": ["Actor", "Critic"],
+            "model_args": ((), ()),
+            "model_kwargs": ({}, {}),
+            "optimizer": "Adam",
+            "criterion": "MSELoss",
+            "criterion_args": (),
+            "criterion_kwargs": {},
+            "lr_scheduler": None,
+            "lr_scheduler_args": None,
+            "lr_scheduler_kwargs": None,
+            "batch_size": 100,
+            "actor_update_times": 5,
+            "critic_update_times": 10,
+            "actor_learning_rate": 0.001,
+            "critic_learning_rate": 0.001,
+            "entropy_weight": None,
+            "value_weight": 0.5,
+            "gradient_max": 1e9,
+            "gae_lambda": 1.0,
+            "discount": 0.99,
+            "normalize_advantage": True,
+            "replay_size": 500000,
+            "replay_device": "cpu",
+            "visualize": False,
+            "visualize_dir": "",
+        }
+        config = config or {}
+        data = config.data if isinstance(config, Config) else dict(config)
+        frame_config = dict(default)
+        frame_config.update(data.get("frame_config", {}))
+        data["frame"] = frame_config["frame"]
+        data["frame_config"] = frame_config
+        return Config(**data)
+
+    @classmethod
+    def init_from_config(cls, config, model_device="cpu"):
+        data = config.data if isinstance(config, Config) else dict(config)
+        fc = data["frame_config"]
+        model_cls = assert_and_get_valid_models(fc["models"])
+        n = len(model_cls)
+        models = [
+            m(*args, **kwargs).to(model_device)
+            for m, args, kwargs in zip(
+                model_cls, fc.get("model_args", ((),) * n),
+                fc.get("model_kwargs", ({},) * n),
+            )
+        ]
+        optimizer = assert_and_get_valid_optimizer(fc["optimizer"])
+        criterion = assert_and_get_valid_criterion(fc["criterion"])(
+            *fc.get("criterion_args", ()), **fc.get("criterion_kwargs", {})
+        )
+        lr_scheduler = assert_and_get_valid_lr_scheduler(fc.get("lr_scheduler"))
+        return cls(
+            *models,
+            optimizer=optimizer,
+            criterion=criterion,
+            lr_scheduler=lr_scheduler,
+            **{
+                k: v
+                for k, v in fc.items()
+                if k
+                not in (
+                    "frame", "models", "model_args", "model_kwargs",
+                    "optimizer", "criterion", "criterion_args",
+                    "criterion_kwargs", "lr_scheduler",
+                )
+            },
+        )

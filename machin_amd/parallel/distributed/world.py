@@ -528,6 +528,15 @@ class RemoteValue:
     local_value = to_here
 
 
+def _rebuild_rpc_group(group_name: str, members: List[str]) -> "RpcGroup":
+    world = get_world()
+    if world is None:
+        raise RuntimeError(
+            "Cannot deserialize an RpcGroup: local world not initialized."
+        )
+    return world.create_rpc_group(group_name, members)
+
+
 class RpcGroup:
     """Name-addressed control-plane RPC within a member set."""
 
@@ -536,6 +545,11 @@ class RpcGroup:
         self.group_name = group_name
         self.members = members
         self.destroyed = False
+
+    def __reduce__(self):
+        # groups cross process boundaries as (name, members) handles
+        # and rebind to the receiving process's world
+        return (_rebuild_rpc_group, (self.group_name, self.members))
 
     # -- membership ----------------------------------------------------
     def size(self) -> int:

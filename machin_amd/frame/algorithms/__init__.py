@@ -11,8 +11,17 @@ from .a2c import A2C
 from .ppo import PPO
 from .trpo import TRPO
 from .gail import GAIL
+from .a3c import A3C
+from .apex import DDPGApex, DQNApex
+from .impala import IMPALA
+from .ars import ARS
 
 __all__ = [
+    "A3C",
+    "DQNApex",
+    "DDPGApex",
+    "IMPALA",
+    "ARS",
     "TorchFramework",
     "DQN",
     "DQNPer",

@@ -1,0 +1,454 @@
+"""IMPALA: importance-weighted actor-learner with V-trace.
+
+Parity target: reference ``machin/frame/algorithms/impala.py``:
+episode-granular distributed buffer (:28-66), stored behavior
+``action_log_prob`` (:217-240), V-trace targets (:317-371) — computed
+here by the gfx950 kernel ``machin_amd.ops.vtrace`` over a padded
+[T, B] batch (the reference runs a per-episode python recursion),
+learner model push to a PushPullModelServer after update.
+
+Padding scheme: an episode of length L < T gets its bootstrap value
+written into ``values[L]`` AND ``rewards[L]`` so the V-trace recursion
+carries vs[L] = bootstrap across the boundary with zero delta; padded
+steps have non-terminal mask 0 and are excluded from losses.
+"""
+import threading
+from typing import Dict, List, Tuple, Union
+
+import numpy as np
+import torch as t
+import torch.nn as nn
+
+from ... import ops
+from ...parallel.distributed.world import RpcGroup
+from ...parallel.server.param_server import PushPullModelServer
+from ...utils.conf import Config
+from ..transition import Transition
+from .base import TorchFramework
+from .utils import (
+    assert_and_get_valid_criterion,
+    assert_and_get_valid_lr_scheduler,
+    assert_and_get_valid_models,
+    assert_and_get_valid_optimizer,
+    safe_call,
+    safe_return,
+)
+
+
+class IMPALABuffer:
+    """Distributed episode queue: members append whole episodes; the
+    learner pops them (consumed once, on-policy-ish)."""
+
+    def __init__(self, buffer_name: str, group: RpcGroup,
+                 buffer_size: int = 500):
+        self.buffer_name = buffer_name
+        self.group = group
+        self.buffer_size = buffer_size
+        self._episodes: List[List[Transition]] = []
+        self._lock = threading.Lock()
+        me = group.get_cur_name()
+        group.register(f"{buffer_name}/{me}/_pop_service",
+                       self._pop_service)
+        group.register(f"{buffer_name}/{me}/_size_service",
+                       self._size_service)
+
+    def store_episode(self, episode: List[Transition]):
+        with self._lock:
+            if len(self._episodes) >= self.buffer_size:
+                self._episodes.pop(0)
+            self._episodes.append(episode)
+
+    def _pop_service(self, max_episodes: int):
+        with self._lock:
+            out = self._episodes[:max_episodes]
+            del self._episodes[:max_episodes]
+            return out
+
+    def _size_service(self) -> int:
+        with self._lock:
+            return len(self._episodes)
+
+    def size(self) -> int:
+        return self._size_service()
+
+    def all_size(self) -> int:
+        futures = [
+            self.group.registered_async(
+                f"{self.buffer_name}/{m}/_size_service"
+            )
+            for m in self.group.get_group_members()
+        ]
+        return sum(f.wait() for f in futures)
+
+    def pop_episodes(self, max_episodes: int) -> List[List[Transition]]:
+        """Pop up to ``max_episodes`` episodes from all members."""
+        out = []
+        for m in self.group.get_group_members():
+            if len(out) >= max_episodes:
+                break
+            got = self.group.registered_sync(
+                f"{self.buffer_name}/{m}/_pop_service",
+                args=(max_episodes - len(out),),
+            )
+            out.extend(got)
+        return out
+
+
+class IMPALA(TorchFramework):
+    _is_top = ["actor", "critic"]
+    _is_restorable = ["actor", "critic"]
+
+    def __init__(
+        self,
+        actor: nn.Module,
+        critic: nn.Module,
+        optimizer,
+        criterion,
+        impala_group: RpcGroup,
+        model_server: Tuple[PushPullModelServer],
+        *_,
+        lr_scheduler=None,
+        lr_scheduler_args=(),
+        lr_scheduler_kwargs=(),
+        batch_size: int = 5,
+        learning_rate: float = 0.001,
+        isw_clip_c: float = 1.0,
+        isw_clip_rho: float = 1.0,
+        entropy_weight: float = None,
+        value_weight: float = 0.5,
+        gradient_max: float = np.inf,
+        discount: float = 0.99,
+        replay_size: int = 500,
+        **__,
+    ):
+        super().__init__()
+        self.batch_size = batch_size
+        self.discount = discount
+        self.value_weight = value_weight
+        self.entropy_weight = entropy_weight
+        self.grad_max = gradient_max
+        self.isw_clip_c = isw_clip_c
+        self.isw_clip_rho = isw_clip_rho
+        self.visualize = False
+
+        self.impala_group = impala_group
+        self.actor = actor
+        self.critic = critic
+        if isinstance(optimizer, (tuple, list)):
+            a_opt, c_opt = optimizer
+        else:
+            a_opt = c_opt = optimizer
+        self.actor_optim = a_opt(actor.parameters(), lr=learning_rate)
+        self.critic_optim = c_opt(critic.parameters(), lr=learning_rate)
+        self.replay_buffer = IMPALABuffer(
+            "impala_buffer", impala_group, replay_size
+        )
+        self.actor_model_server = model_server[0]
+        self.is_syncing = True
+
+        self.actor_lr_sch = None
+        self.critic_lr_sch = None
+        if lr_scheduler is not None:
+            args = lr_scheduler_args or ((), ())
+            kwargs = lr_scheduler_kwargs or ({}, {})
+            self.actor_lr_sch = lr_scheduler(self.actor_optim, *args[0],
+                                             **kwargs[0])
+            self.critic_lr_sch = lr_scheduler(self.critic_optim, *args[1],
+                                              **kwargs[1])
+        self.criterion = (
+            criterion() if isinstance(criterion, type) else criterion
+        )
+
+    @property
+    def optimizers(self):
+        return [self.actor_optim, self.critic_optim]
+
+    @optimizers.setter
+    def optimizers(self, optimizers):
+        self.actor_optim, self.critic_optim = optimizers
+
+    @property
+    def lr_schedulers(self):
+        out = []
+        if self.actor_lr_sch is not None:
+            out.append(self.actor_lr_sch)
+        if self.critic_lr_sch is not None:
+            out.append(self.critic_lr_sch)
+        return out
+
+    @classmethod
+    def is_distributed(cls) -> bool:
+        return True
+
+    def set_sync(self, is_syncing):
+        self.is_syncing = is_syncing
+
+    def manual_sync(self):
+        self.actor_model_server.pull(self.actor)
+
+    # ------------------------------------------------------------------
+    def act(self, state: Dict, *_, **__):
+        """Sample action: returns actor output tuple
+        (action, log_prob[, entropy])."""
+        if self.is_syncing:
+            self.actor_model_server.pull(self.actor)
+        return safe_call(self.actor, state)
+
+    def _eval_act(self, state: Dict, action: Dict, **__):
+        return safe_call(self.actor, state, action)
+
+    def _criticize(self, state: Dict, **__):
+        return safe_return(safe_call(self.critic, state))
+
+    # ------------------------------------------------------------------
+    def store_transition(self, transition):
+        raise NotImplementedError(
+            "IMPALA requires whole episodes: use store_episode()."
+        )
+
+    def store_episode(self, episode: List[Union[Transition, Dict]]):
+        """Store one episode; each transition must carry the behavior
+        policy's ``action_log_prob`` custom attribute."""
+        transitions = []
+        for tr in episode:
+            if isinstance(tr, dict):
+                tr = Transition(**tr)
+            if "action_log_prob" not in tr.keys():
+                raise ValueError(
+                    "IMPALA transitions require 'action_log_prob'."
+                )
+            transitions.append(tr._detach().to("cpu"))
+        self.replay_buffer.store_episode(transitions)
+
+    # ------------------------------------------------------------------
+    def update(self, update_value=True, update_target=True,
+               concatenate_samples=True, **__):
+        """Learner: pop episodes, compute V-trace targets, update
+        actor + critic, push the actor to the model server."""
+        episodes = self.replay_buffer.pop_episodes(self.batch_size)
+        if not episodes:
+            return 0.0, 0.0
+        self.actor.train()
+        self.critic.train()
+
+        B = len(episodes)
+        lengths = [len(ep) for ep in episodes]
+        T = max(lengths)
+
+        # flatten all real steps for one batched forward
+        state_keys = episodes[0][0].state.keys()
+        flat_state = {
+            k: t.cat(
+                [tr.state[k] for ep in episodes for tr in ep], dim=0
+            )
+            for k in state_keys
+        }
+        flat_action = {
+            k: t.cat(
+                [tr.action[k] for ep in episodes for tr in ep], dim=0
+            )
+            for k in episodes[0][0].action.keys()
+        }
+        last_next_state = {
+            k: t.cat([ep[-1].next_state[k] for ep in episodes], dim=0)
+            for k in state_keys
+        }
+
+        result = self._eval_act(flat_state, flat_action)
+        taken_logp_flat = result[1].view(-1)
+        entropy_flat = (
+            result[2].view(-1) if len(result) > 2 else None
+        )
+        values_flat = self._criticize(flat_state).view(-1)
+        device = values_flat.device
+
+        with t.no_grad():
+            boot_all = self._criticize(last_next_state).view(B)
+            last_terminal = t.tensor(
+                [float(self._scalar(ep[-1].terminal)) for ep in episodes],
+                device=device,
+            )
+            boot_all = boot_all * (1.0 - last_terminal)
+
+        # scatter into padded [T, B]
+        mask = t.zeros(T, B, device=device)
+        rewards = t.zeros(T, B, device=device)
+        nd = t.zeros(T, B, device=device)
+        blp = t.zeros(T, B, device=device)
+        tlp_pad = t.zeros(T, B, device=device)
+        values_pad = t.zeros(T, B, device=device)
+        idx = 0
+        for b, ep in enumerate(episodes):
+            L = lengths[b]
+            mask[:L, b] = 1.0
+            for s, tr in enumerate(ep):
+                rewards[s, b] = float(self._scalar(tr.reward))
+                nd[s, b] = 1.0 - float(self._scalar(tr.terminal))
+                blp[s, b] = float(self._scalar(tr.action_log_prob))
+            # bootstrap carry: values[L] = rewards[L] = boot
+            if L < T:
+                values_pad[L, b] = boot_all[b]
+                rewards[L, b] = boot_all[b]
+            idx += L
+
+        # scatter network outputs (keep autograd through values/logp)
+        pos = t.cat(
+            [
+                t.arange(L, device=device) * B + b
+                for b, L in enumerate(lengths)
+            ]
+        )
+        tlp_pad = tlp_pad.view(-1).index_put(
+            (pos,), taken_logp_flat
+        ).view(T, B)
+        values_pad = values_pad.view(-1).index_put(
+            (pos,), values_flat, accumulate=True
+        ).view(T, B)
+
+        with t.no_grad():
+            vs, pg_adv = ops.vtrace(
+                blp,
+                tlp_pad.detach(),
+                rewards,
+                values_pad.detach(),
+                boot_all,
+                nd * mask,
+                self.discount,
+                rho_clip=self.isw_clip_rho,
+                c_clip=self.isw_clip_c,
+                pg_rho_clip=self.isw_clip_rho,
+            )
+
+        n_valid = mask.sum().clamp_min(1.0)
+        act_policy_loss = -(pg_adv * tlp_pad * mask).sum() / n_valid
+        if self.entropy_weight is not None and entropy_flat is not None:
+            act_policy_loss = (
+                act_policy_loss
+                - self.entropy_weight * entropy_flat.mean()
+            )
+        value_loss = (
+            self.criterion(values_pad * mask, vs * mask) / n_valid
+        ) * self.value_weight
+
+        if update_value:
+            self.critic_optim.zero_grad(set_to_none=True)
+            self._backward(value_loss, retain_graph=True)
+            nn.utils.clip_grad_norm_(self.critic.parameters(), self.grad_max)
+            self.critic_optim.step()
+        self.actor_optim.zero_grad(set_to_none=True)
+        self._backward(act_policy_loss)
+        nn.utils.clip_grad_norm_(self.actor.parameters(), self.grad_max)
+        self.actor_optim.step()
+
+        if update_target:
+            self.actor_model_server.push(
+                getattr(self.actor, "module", self.actor)
+            )
+        return (
+            -float(act_policy_loss.detach().item()),
+            float(value_loss.detach().item()),
+        )
+
+    @staticmethod
+    def _scalar(v):
+        return v.reshape(-1)[0].item() if t.is_tensor(v) else v
+
+    def update_lr_scheduler(self):
+        if self.actor_lr_sch is not None:
+            self.actor_lr_sch.step()
+        if self.critic_lr_sch is not None:
+            self.critic_lr_sch.step()
+
+    # ------------------------------------------------------------------
+    @classmethod
+    def generate_config(cls, config):
+        default = {
+            "frame": "IMPALA",
+            "models": ["Actor", "Critic"],
+            "model_args": ((), ()),
+            "model_kwargs": ({}, {}),
+            "optimizer": "Adam",
+            "criterion": "MSELoss",
+            "criterion_args": (),
+            "criterion_kwargs": {"reduction": "sum"},
+            "lr_scheduler": None,
+            "batch_size": 5,
+            "learning_rate": 0.001,
+            "isw_clip_c": 1.0,
+            "isw_clip_rho": 1.0,
+            "entropy_weight": None,
+            "value_weight": 0.5,
+            "gradient_max": 1e9,
+            "discount": 0.99,
+            "replay_size": 500,
+            "impala_group_name": "impala_group",
+            "impala_group_members": "all",
+            "learner_process_number": 1,
+        }
+        config = config or {}
+        data = config.data if isinstance(config, Config) else dict(config)
+        frame_config = dict(default)
+        frame_config.update(data.get("frame_config", {}))
+        data["frame"] = frame_config["frame"]
+        data["frame_config"] = frame_config
+        return Config(**data)
+
+    @classmethod
+    def init_from_config(cls, config, model_device="cpu"):
+        from ...frame.helpers.servers import model_server_helper
+        from ...parallel.ddp import DistributedDataParallel
+        from ...parallel.distributed.world import get_world
+
+        data = config.data if isinstance(config, Config) else dict(config)
+        fc = data["frame_config"]
+        world = get_world()
+        servers = model_server_helper(model_num=1)
+        members = (
+            world.get_members()
+            if fc.get("impala_group_members", "all") == "all"
+            else fc["impala_group_members"]
+        )
+        group = world.create_rpc_group(
+            fc.get("impala_group_name", "impala_group"), members
+        )
+        model_cls = assert_and_get_valid_models(fc["models"])
+        models = [
+            m(*args, **kwargs).to(model_device)
+            for m, args, kwargs in zip(
+                model_cls, fc.get("model_args", ((), ())),
+                fc.get("model_kwargs", ({}, {})),
+            )
+        ]
+        learner_n = fc.get("learner_process_number", 1)
+        learner_ranks = list(range(learner_n))
+        if world.rank in learner_ranks and learner_n > 1:
+            coll = world.create_collective_group(learner_ranks)
+            models[0] = DistributedDataParallel(
+                models[0], process_group=coll.group
+            )
+        optimizer = assert_and_get_valid_optimizer(fc["optimizer"])
+        criterion = assert_and_get_valid_criterion(fc["criterion"])(
+            *fc.get("criterion_args", ()), **fc.get("criterion_kwargs", {})
+        )
+        lr_scheduler = assert_and_get_valid_lr_scheduler(
+            fc.get("lr_scheduler")
+        )
+        frame = cls(
+            models[0], models[1], optimizer, criterion, group, servers,
+            lr_scheduler=lr_scheduler,
+            **{
+                k: v
+                for k, v in fc.items()
+                if k
+                not in (
+                    "frame", "models", "model_args", "model_kwargs",
+                    "optimizer", "criterion", "criterion_args",
+                    "criterion_kwargs", "lr_scheduler",
+                    "impala_group_name", "impala_group_members",
+                    "learner_process_number",
+                )
+            },
+        )
+        if world.rank not in learner_ranks:
+            frame.update = lambda *a, **k: (0.0, 0.0)
+        return frame

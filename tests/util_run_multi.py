@@ -20,6 +20,9 @@ def free_port() -> int:
 def _worker(rank, world_size, port, func_bytes, args, result_queue,
             use_world, names):
     try:
+        import torch
+
+        torch.set_num_threads(2)
         os.environ["MASTER_ADDR"] = "127.0.0.1"
         os.environ["MASTER_PORT"] = str(port)
         os.environ["GLOO_SOCKET_IFNAME"] = os.environ.get(
@@ -52,7 +55,9 @@ def run_multi(func, world_size: int = 3, args=(), timeout: float = 120,
               use_world: bool = True, names=None):
     """Run ``func(rank[, world], *args)`` in ``world_size`` processes.
     Returns the list of per-rank results, ordered by rank."""
-    ctx = mp.get_context("fork")
+    # spawn, not fork: forking a test process with a warm OpenMP/torch
+    # runtime intermittently deadlocks the children
+    ctx = mp.get_context("spawn")
     port = free_port()
     rq = ctx.Queue()
     procs = [

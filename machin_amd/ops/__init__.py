@@ -100,7 +100,7 @@ def discounted_returns(
     if rewards.is_cuda:
         ext = _require_ext()
         out = ext.discounted_returns(
-            rewards.contiguous(), terminals.contiguous(),
+            rewards.contiguous(), (1.0 - terminals).contiguous(),
             bootstrap.contiguous().float(), float(gamma),
         )
     else:
@@ -141,7 +141,7 @@ def gae(
         ext = _require_ext()
         out = ext.gae(
             rewards.contiguous(), values.contiguous(),
-            next_values.contiguous(), terminals.contiguous(),
+            next_values.contiguous(), (1.0 - terminals).contiguous(),
             float(gamma), float(lam),
         )
     else:

@@ -36,3 +36,6 @@ __all__ = [
     "TRPO",
     "GAIL",
 ]
+from .maddpg import MADDPG  # noqa: E402
+
+__all__.append("MADDPG")

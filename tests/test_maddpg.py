@@ -1,0 +1,121 @@
+"""MADDPG API tests."""
+import pytest
+import torch as t
+import torch.nn as nn
+
+from machin_amd.frame.algorithms import MADDPG
+
+
+class Actor(nn.Module):
+    def __init__(self, state_dim=3, action_dim=1):
+        super().__init__()
+        self.fc = nn.Linear(state_dim, action_dim)
+
+    def forward(self, state):
+        return t.tanh(self.fc(state))
+
+
+class Critic(nn.Module):
+    """Centralized: sees the concatenation of visible agents'
+    states and actions."""
+
+    def __init__(self, n_vis=3, state_dim=3, action_dim=1):
+        super().__init__()
+        self.fc = nn.Linear(n_vis * (state_dim + action_dim), 1)
+
+    def forward(self, state, action):
+        return self.fc(t.cat([state, action], dim=1))
+
+
+def make_maddpg(n=3, sub_policy_num=0, **kwargs):
+    return MADDPG(
+        [Actor() for _ in range(n)],
+        [Actor() for _ in range(n)],
+        [Critic(n) for _ in range(n)],
+        [Critic(n) for _ in range(n)],
+        t.optim.Adam,
+        nn.MSELoss(),
+        batch_size=8,
+        sub_policy_num=sub_policy_num,
+        **kwargs,
+    )
+
+
+def make_episodes(n=3, length=10):
+    return [
+        [
+            {
+                "state": {"state": t.rand(1, 3)},
+                "action": {"action": t.rand(1, 1)},
+                "next_state": {"state": t.rand(1, 3)},
+                "reward": 0.5,
+                "terminal": i == length - 1,
+            }
+            for i in range(length)
+        ]
+        for _ in range(n)
+    ]
+
+
+class TestMADDPG:
+    def test_act(self):
+        m = make_maddpg()
+        states = [{"state": t.rand(1, 3)} for _ in range(3)]
+        acts = m.act(states)
+        assert len(acts) == 3 and acts[0].shape == (1, 1)
+        acts = m.act_with_noise(states, noise_param=(0.0, 0.1),
+                                mode="normal")
+        assert len(acts) == 3
+        with pytest.raises(ValueError):
+            m.act_with_noise(states, mode="bogus")
+
+    def test_update(self):
+        m = make_maddpg(sub_policy_num=1)
+        m.store_episodes(make_episodes())
+        pl, vl = m.update()
+        assert isinstance(pl, float) and isinstance(vl, float)
+
+    def test_visibility_matrix(self):
+        n = 3
+
+        class PartialCritic(nn.Module):
+            def __init__(self):
+                super().__init__()
+                self.fc = nn.Linear(2 * (3 + 1), 1)
+
+            def forward(self, state, action):
+                return self.fc(t.cat([state, action], dim=1))
+
+        m = MADDPG(
+            [Actor() for _ in range(n)],
+            [Actor() for _ in range(n)],
+            [PartialCritic() for _ in range(n)],
+            [PartialCritic() for _ in range(n)],
+            t.optim.Adam,
+            nn.MSELoss(),
+            batch_size=8,
+            critic_visible_actors=[[0, 1], [1, 2], [2, 0]],
+        )
+        m.store_episodes(make_episodes())
+        pl, vl = m.update()
+        assert pl == pl
+
+    def test_lockstep_validation(self):
+        m = make_maddpg()
+        eps = make_episodes()
+        eps[1] = eps[1][:5]
+        with pytest.raises(ValueError):
+            m.store_episodes(eps)
+
+    def test_save_load(self, tmp_path):
+        m = make_maddpg()
+        m.store_episodes(make_episodes())
+        m.update()
+        m.save(str(tmp_path))
+        m2 = make_maddpg()
+        m2.load(str(tmp_path))
+        for p1, p2 in zip(
+            m.all_actor_target.parameters(),
+            m2.all_actor_target.parameters(),
+        ):
+            assert t.allclose(p1, p2)

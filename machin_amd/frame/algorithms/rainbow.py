@@ -184,3 +184,58 @@ class RAINBOW(DQNPer):
                 if self._update_counter % self.update_steps == 0:
                     hard_update(self.qnet_target, self.qnet)
         return float(loss.detach().item())
+
+    # ------------------------------------------------------------------
+    @classmethod
+    def generate_config(cls, config):
+        from .dqn import DQN
+
+        config = DQN.generate_config(config)
+        config["frame"] = "RAINBOW"
+        fc = config["frame_config"]
+        fc["frame"] = "RAINBOW"
+        fc.pop("mode", None)
+        fc.setdefault("value_min", -10.0)
+        fc.setdefault("value_max", 10.0)
+        fc.setdefault("reward_future_steps", 3)
+        return config
+
+    @classmethod
+    def init_from_config(cls, config, model_device="cpu"):
+        from ...utils.conf import Config
+        from .utils import (
+            assert_and_get_valid_criterion,
+            assert_and_get_valid_models,
+            assert_and_get_valid_optimizer,
+        )
+
+        data = config.data if isinstance(config, Config) else dict(config)
+        fc = data["frame_config"]
+        model_cls = assert_and_get_valid_models(fc["models"])
+        models = [
+            m(*args, **kwargs).to(model_device)
+            for m, args, kwargs in zip(
+                model_cls, fc.get("model_args", ((), ())),
+                fc.get("model_kwargs", ({}, {})),
+            )
+        ]
+        optimizer = assert_and_get_valid_optimizer(fc["optimizer"])
+        criterion = assert_and_get_valid_criterion(fc["criterion"])(
+            *fc.get("criterion_args", ()), **fc.get("criterion_kwargs", {})
+        )
+        return cls(
+            models[0], models[1], optimizer,
+            fc["value_min"], fc["value_max"],
+            criterion=criterion,
+            **{
+                k: v
+                for k, v in fc.items()
+                if k
+                not in (
+                    "frame", "models", "model_args", "model_kwargs",
+                    "optimizer", "criterion", "criterion_args",
+                    "criterion_kwargs", "lr_scheduler", "value_min",
+                    "value_max", "mode",
+                )
+            },
+        )

@@ -1,0 +1,3 @@
+from .classic_control import CartPoleEnv, PendulumEnv, Space, make
+
+__all__ = ["CartPoleEnv", "PendulumEnv", "Space", "make"]

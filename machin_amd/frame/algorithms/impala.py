@@ -312,7 +312,7 @@ class IMPALA(TorchFramework):
                 rewards,
                 values_pad.detach(),
                 boot_all,
-                nd * mask,
+                1.0 - nd * mask,  # terminals: padding counts as terminal
                 self.discount,
                 rho_clip=self.isw_clip_rho,
                 c_clip=self.isw_clip_c,

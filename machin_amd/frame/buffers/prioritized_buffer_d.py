@@ -63,7 +63,11 @@ class DistributedPrioritizedBuffer(PrioritizedBuffer):
             self._entry_versions[:] = 0
         return True
 
-    def _sample_service(self, batch_size: int, all_weight_sum: float):
+    def _sample_service(self, batch_size: int, all_weight_sum: float,
+                        sample_attrs=None,
+                        additional_concat_custom_attrs=None):
+        """Member-side: sample by priority AND concatenate; replies
+        carry flat per-attribute tensors (see buffer_d)."""
         with self.wr_lock:
             if len(self.storage) == 0 or batch_size <= 0:
                 return 0, None, None, None, None
@@ -72,7 +76,11 @@ class DistributedPrioritizedBuffer(PrioritizedBuffer):
             )
             batch = [self.storage[int(i)] for i in index]
             versions = self._entry_versions[index].copy()
-            return batch_size, batch, index, is_weight, versions
+            result = self.post_process_batch(
+                batch, "cpu", True, sample_attrs,
+                additional_concat_custom_attrs,
+            )
+            return batch_size, result, index, is_weight, versions
 
     def _update_priority_service(self, priorities, indexes, versions):
         with self.wr_lock:
@@ -157,31 +165,33 @@ class DistributedPrioritizedBuffer(PrioritizedBuffer):
                 m,
                 self.group.registered_async(
                     f"{self.buffer_name}/{m}/_sample_service",
-                    args=(c, all_weight_sum),
+                    args=(c, all_weight_sum, sample_attrs,
+                          additional_concat_custom_attrs),
                 ),
             )
             for m, c in zip(members, counts)
             if c > 0
         ]
-        all_batch = []
+        parts = []
         all_index = {}
         all_is_weight = []
+        total = 0
         for m, f in sample_futures:
-            bsize, batch, index, is_weight, versions = f.wait()
+            bsize, result, index, is_weight, versions = f.wait()
             if bsize > 0:
-                all_batch.extend(batch)
+                parts.append(result)
+                total += bsize
                 all_index[m] = (index, versions)
                 all_is_weight.append(is_weight)
-        if not all_batch:
+        if not parts:
             return 0, None, None, None
         is_weight = np.concatenate(all_is_weight)
         # renormalize IS weights globally
         is_weight = is_weight / max(is_weight.max(), 1e-12)
-        result = self.post_process_batch(
-            all_batch, device, concatenate, sample_attrs,
-            additional_concat_custom_attrs,
-        )
-        return len(all_batch), result, all_index, is_weight
+        from .buffer_d import merge_processed_batches
+
+        result = merge_processed_batches(parts, device)
+        return total, result, all_index, is_weight
 
     def update_priority(self, priorities: np.ndarray, indexes: dict):
         """Route new priorities back to each owning member;

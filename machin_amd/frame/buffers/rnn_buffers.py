@@ -227,7 +227,9 @@ class RNNDistributedPrioritizedBuffer(
             )
         return handles
 
-    def _sample_service(self, batch_size: int, all_weight_sum: float):
+    def _sample_service(self, batch_size: int, all_weight_sum: float,
+                        sample_attrs=None,
+                        additional_concat_custom_attrs=None):
         with self.wr_lock:
             if len(self.storage) == 0 or batch_size <= 0:
                 return 0, None, None, None, None
@@ -248,11 +250,17 @@ class RNNDistributedPrioritizedBuffer(
                 batch.extend(self._window(ep, off))
                 kept_index.append(int(i))
                 kept_weight.append(w)
+            if not batch:
+                return 0, None, None, None, None
             versions = self._entry_versions[np.asarray(kept_index,
                                                        dtype=np.int64)]
+            result = self.post_process_batch(
+                batch, "cpu", True, sample_attrs,
+                additional_concat_custom_attrs,
+            )
             return (
                 len(kept_index),
-                batch,
+                result,
                 np.asarray(kept_index),
                 np.asarray(kept_weight),
                 versions,

@@ -329,10 +329,13 @@ class World:
             with self._lut_lock:
                 return request["key"] in self._lut
         if op == "barrier_enter":
-            return self._barrier_enter(request["key"], request["count"])
+            return self._barrier_enter(
+                request["key"], request["count"],
+                request.get("timeout", self.rpc_timeout),
+            )
         raise ValueError(f"Unknown control op {op!r}")
 
-    def _barrier_enter(self, key, count):
+    def _barrier_enter(self, key, count, timeout=None):
         with self._barriers_lock:
             state = self._barriers.get(key)
             if state is None:
@@ -343,7 +346,7 @@ class World:
                 state["event"].set()
                 del self._barriers[key]
             event = state["event"]
-        if not event.wait(self.rpc_timeout):
+        if not event.wait(timeout or self.rpc_timeout):
             raise TimeoutError(f"Barrier {key!r} timed out.")
         return True
 
@@ -669,16 +672,40 @@ class RpcGroup:
 
     def registered_sync(self, key, args=(), kwargs=None,
                         timeout: float = None) -> Any:
-        holder = self.world._control(
-            self.world.lut_manager,
-            {"op": "lut_get", "key": ("s", self._lut_key(key))},
-        )
-        return self.world._control(
-            holder,
-            {"op": "call_service", "key": self._lut_key(key), "args": args,
-             "kwargs": kwargs or {}},
-            timeout,
-        )
+        # service holders are stable: cache the LUT lookup, retry once
+        # on failure in case the service moved / was re-registered
+        cache = self.__dict__.setdefault("_service_lut_cache", {})
+        holder = cache.get(key)
+        if holder is None:
+            holder = self.world._control(
+                self.world.lut_manager,
+                {"op": "lut_get", "key": ("s", self._lut_key(key))},
+            )
+            cache[key] = holder
+        try:
+            return self.world._control(
+                holder,
+                {"op": "call_service", "key": self._lut_key(key),
+                 "args": args, "kwargs": kwargs or {}},
+                timeout,
+            )
+        except RuntimeError as e:
+            # retry ONLY for "service moved" — a service that RAISED
+            # must not be re-executed
+            if "not registered here" not in str(e):
+                raise
+            cache.pop(key, None)
+            holder = self.world._control(
+                self.world.lut_manager,
+                {"op": "lut_get", "key": ("s", self._lut_key(key))},
+            )
+            cache[key] = holder
+            return self.world._control(
+                holder,
+                {"op": "call_service", "key": self._lut_key(key),
+                 "args": args, "kwargs": kwargs or {}},
+                timeout,
+            )
 
     def registered_async(self, key, args=(), kwargs=None,
                          timeout: float = None) -> _Future:
@@ -702,7 +729,10 @@ class RpcGroup:
     _barrier_gen = 0
 
     def barrier(self, timeout: float = None):
-        """Group-wide barrier through the leader (first member)."""
+        """Group-wide barrier through the leader (first member).
+        Default timeout is long (600 s): members may arrive minutes
+        apart during training phases."""
+        timeout = timeout if timeout is not None else 600.0
         self._barrier_gen += 1
         leader = self.members[0]
         self.world._control(
@@ -711,8 +741,9 @@ class RpcGroup:
                 "op": "barrier_enter",
                 "key": (self.group_name, self._barrier_gen),
                 "count": len(self.members),
+                "timeout": timeout,
             },
-            timeout or self.world.rpc_timeout,
+            timeout + 10.0,
         )
 
     def destroy(self):

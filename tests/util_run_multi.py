@@ -71,12 +71,19 @@ def run_multi(func, world_size: int = 3, args=(), timeout: float = 120,
     for p in procs:
         p.start()
     results = {}
+    failures = {}
     try:
         for _ in range(world_size):
             rank, ok, value = rq.get(timeout=timeout)
-            if not ok:
-                raise AssertionError(f"Rank {rank} failed:\n{value}")
-            results[rank] = value
+            if ok:
+                results[rank] = value
+            else:
+                failures[rank] = value
+        if failures:
+            report = "\n\n".join(
+                f"Rank {r} failed:\n{v}" for r, v in sorted(failures.items())
+            )
+            raise AssertionError(report)
     finally:
         for p in procs:
             p.join(timeout=10)

@@ -42,6 +42,7 @@ class ImpalaLearnerBench:
         value_weight: float = 0.5,
         discount: float = 0.99,
         grad_clip: float = 40.0,
+        capturable: bool = False,
     ):
         from machin_amd.model.nets.nature_cnn import ActorCriticCNN
         import machin_amd.ops as ops
@@ -72,7 +73,8 @@ class ImpalaLearnerBench:
                 for p in self.model.parameters():
                     dist.broadcast(p.data, src=0)
         self.optim = t.optim.RMSprop(
-            self.model.parameters(), lr=lr, alpha=0.99, eps=0.1
+            self.model.parameters(), lr=lr, alpha=0.99, eps=0.1,
+            capturable=capturable,
         )
 
         # synthetic rollout pool (uint8 frames like a real Atari actor
@@ -218,6 +220,7 @@ def main():
         env_batch=args.env_batch,
         action_num=args.actions,
         distributed=distributed,
+        capturable=args.graph and not distributed,
     )
 
     if args.graph and not distributed:

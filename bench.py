@@ -123,7 +123,17 @@ class ImpalaLearnerBench:
     def _step_body(self, data):
         T, B = self.unroll, self.env_batch
 
-        frames = data["frames"].to(self.dtype).mul_(1.0 / 255.0)
+        if self.dtype == t.bfloat16:
+            # fused u8->bf16 dequant (gfx950 kernel): one vectorized
+            # pass instead of .to() + .mul_(). The pool frames are
+            # channels_last; permute exposes their linear memory so
+            # the kernel runs on the raw bytes and the result keeps
+            # the NHWC stride order MIOpen wants.
+            nhwc = data["frames"].permute(0, 2, 3, 1)
+            out = self.ops.dequant_u8(nhwc, 1.0 / 255.0)
+            frames = out.view(nhwc.shape).permute(0, 3, 1, 2)
+        else:
+            frames = data["frames"].to(self.dtype).mul_(1.0 / 255.0)
         with t.autocast(device_type="cuda", dtype=self.dtype):
             logits, values = self.model(frames)
         logits = logits.float().view(T, B, -1)

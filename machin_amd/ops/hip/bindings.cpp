@@ -36,6 +36,8 @@ void normal_noise_launch(float*, int64_t, float, float, uint64_t, uint64_t,
                          int, hipStream_t);
 void ou_update_launch(float*, int64_t, float, float, float, float, uint64_t,
                       uint64_t, hipStream_t);
+void u8_to_bf16_scale_launch(const unsigned char*, void*, int64_t, float,
+                             hipStream_t);
 
 namespace {
 
@@ -229,6 +231,17 @@ void ou_update_(Tensor x, double mu, double theta, double sigma, double dt,
                    (uint64_t)offset, current_stream());
 }
 
+Tensor u8_to_bf16_scale(Tensor in, double scale) {
+  TORCH_CHECK(in.is_cuda() && in.scalar_type() == at::kByte &&
+                  in.is_contiguous(),
+              "input must be contiguous uint8 CUDA");
+  const at::cuda::OptionalCUDAGuard guard(in.device());
+  Tensor out = at::empty_like(in, in.options().dtype(at::kBFloat16));
+  u8_to_bf16_scale_launch(in.data_ptr<unsigned char>(), out.data_ptr(),
+                          in.numel(), (float)scale, current_stream());
+  return out;
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -245,4 +258,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gaussian_logprob", &gaussian_logprob);
   m.def("normal_noise_", &normal_noise_);
   m.def("ou_update_", &ou_update_);
+  m.def("u8_to_bf16_scale", &u8_to_bf16_scale);
 }

@@ -24,6 +24,7 @@ ext = CUDAExtension(
         os.path.join(HIP_DIR, "projection.hip"),
         os.path.join(HIP_DIR, "multi_tensor.hip"),
         os.path.join(HIP_DIR, "distributions.hip"),
+        os.path.join(HIP_DIR, "elementwise.hip"),
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],

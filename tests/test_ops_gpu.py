@@ -212,3 +212,22 @@ class TestDistributionsGPU:
         # stationary: mean ~ 0, std ~ sigma/sqrt(2*theta) = 0.365
         assert abs(x.mean().item()) < 0.05
         assert abs(x.std().item() - 0.2 / (2 * 0.15) ** 0.5) < 0.05
+
+
+class TestDequantGPU:
+    def test_u8_to_bf16(self, dev):
+        import machin_amd.ops as ops
+
+        for n in (16 * 1000, 16 * 1000 + 7, 5):
+            x = t.randint(0, 256, (n,), dtype=t.uint8, device=dev)
+            out = ops.dequant_u8(x, 1.0 / 255.0)
+            assert out.dtype == t.bfloat16
+            expect = (x.float() / 255.0).to(t.bfloat16)
+            assert t.allclose(out.float(), expect.float(), atol=1e-3)
+
+    def test_shape_preserved(self, dev):
+        import machin_amd.ops as ops
+
+        x = t.randint(0, 256, (8, 4, 84, 84), dtype=t.uint8, device=dev)
+        out = ops.dequant_u8(x)
+        assert out.shape == x.shape

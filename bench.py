@@ -31,7 +31,7 @@ class ImpalaLearnerBench:
         self,
         device="cuda:0",
         unroll: int = 20,
-        env_batch: int = 256,
+        env_batch: int = 2048,
         action_num: int = 6,
         frames: int = 4,
         dtype=t.bfloat16,
@@ -203,7 +203,7 @@ def main():
     parser.add_argument("--steps", type=int, default=30)
     parser.add_argument("--warmup", type=int, default=10)
     parser.add_argument("--unroll", type=int, default=20)
-    parser.add_argument("--env-batch", type=int, default=256)
+    parser.add_argument("--env-batch", type=int, default=2048)
     parser.add_argument("--actions", type=int, default=6)
     parser.add_argument("--graph", action="store_true",
                         help="capture the learner step in a hipGraph")

@@ -160,10 +160,32 @@ class BasePool:
             if res is not None:
                 res._set(success, value)
 
+    # -- worker maintenance -------------------------------------------
+    def _maintain_workers(self):
+        """Respawn dead workers (reference pool.py:645-665 keeps the
+        pool population stable across worker crashes)."""
+        for i, w in enumerate(self._workers):
+            if not w.is_alive() and not self._closed:
+                ctx = self._ctx
+                new = ctx.Process(
+                    target=_worker_loop,
+                    args=(
+                        self._task_get_fn(i),
+                        self._result_put_fn(i),
+                        self._initializer,
+                        self._initargs,
+                        self._contexts[i] if self._contexts else None,
+                    ),
+                    daemon=True,
+                )
+                new.start()
+                self._workers[i] = new
+
     # -- submission ----------------------------------------------------
     def _submit(self, func, args, kwargs, needs_ctx=False) -> AsyncResult:
         if self._closed:
             raise RuntimeError("Pool is closed.")
+        self._maintain_workers()
         task_id = next(self._counter)
         res = AsyncResult()
         with self._results_lock:

@@ -38,6 +38,9 @@ void ou_update_launch(float*, int64_t, float, float, float, float, uint64_t,
                       uint64_t, hipStream_t);
 void u8_to_bf16_scale_launch(const unsigned char*, void*, int64_t, float,
                              hipStream_t);
+void conv1_wrw_launch(const void*, const unsigned char*, float*, float*,
+                      int64_t, float, hipStream_t);
+void mfma_probe_launch(const void*, const void*, float*, hipStream_t);
 
 namespace {
 
@@ -242,6 +245,41 @@ Tensor u8_to_bf16_scale(Tensor in, double scale) {
   return out;
 }
 
+Tensor conv1_wrw(Tensor dy, Tensor frames, double scale) {
+  // dy: [K, 32] bf16 (NHWC-flattened conv output grad);
+  // frames: [B, 84, 84, 4] u8. Returns [32, 4, 8, 8] fp32 grad.
+  TORCH_CHECK(dy.is_cuda() && dy.scalar_type() == at::kBFloat16 &&
+                  dy.is_contiguous() && dy.size(1) == 32,
+              "dy must be contiguous [K,32] bf16 CUDA");
+  TORCH_CHECK(frames.is_cuda() && frames.scalar_type() == at::kByte &&
+                  frames.is_contiguous(),
+              "frames must be contiguous u8 CUDA");
+  TORCH_CHECK(dy.size(0) == frames.size(0) * 400,
+              "K must equal batch*400");
+  const at::cuda::OptionalCUDAGuard guard(dy.device());
+  Tensor scratch = at::empty({32 * 256},
+                             dy.options().dtype(at::kFloat));
+  Tensor grad_w = at::empty({32, 4, 8, 8},
+                            dy.options().dtype(at::kFloat));
+  conv1_wrw_launch(dy.data_ptr(), frames.data_ptr<unsigned char>(),
+                   scratch.data_ptr<float>(), grad_w.data_ptr<float>(),
+                   dy.size(0), (float)scale, current_stream());
+  return grad_w;
+}
+
+Tensor mfma_probe(Tensor A, Tensor B) {
+  TORCH_CHECK(A.is_cuda() && A.scalar_type() == at::kBFloat16 &&
+              A.is_contiguous() && A.size(0) == 16 && A.size(1) == 32,
+              "A must be [16,32] bf16");
+  TORCH_CHECK(B.is_contiguous() && B.size(0) == 32 && B.size(1) == 16,
+              "B must be [32,16] bf16");
+  const at::cuda::OptionalCUDAGuard guard(A.device());
+  Tensor D = at::empty({16, 16}, A.options().dtype(at::kFloat));
+  mfma_probe_launch(A.data_ptr(), B.data_ptr(), D.data_ptr<float>(),
+                    current_stream());
+  return D;
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -259,4 +297,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("normal_noise_", &normal_noise_);
   m.def("ou_update_", &ou_update_);
   m.def("u8_to_bf16_scale", &u8_to_bf16_scale);
+  m.def("conv1_wrw", &conv1_wrw);
+  m.def("mfma_probe", &mfma_probe);
 }

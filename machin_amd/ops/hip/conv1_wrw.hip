@@ -1,0 +1,216 @@
+// gfx950 fused weight-gradient kernel for the Atari stem conv
+// (Conv2d(C_in=4, C_out=32, kernel 8x8, stride 4) on 84x84 frames).
+//
+// Why hand-written: the weight gradient of this layer is a skinny
+// reduction GEMM dW[32,256] = dy^T[32,K] @ im2col(x)[K,256] with
+// K = batch*400 (16.4M at bench batch 40960). MIOpen/CK pick kernels
+// that run it at ~2% MFMA efficiency (2.5-5.4 ms/step measured,
+// profiles/impala_bench_kernel_trace_r01.md). The op is HBM-bound in
+// principle: dy (1.05 GB bf16) + x (1.16 GB u8, read ONCE as u8 and
+// dequantized in-register) ≈ 2.2 GB -> ~0.4 ms at 6 TB/s.
+//
+// Design (guide §3, §5):
+// - mfma_f32_16x16x32_bf16 tiles; output 32x256 = 2x16 tiles of
+//   16x16; 4 waves per workgroup, each owns (M-tile, 8 N-tiles),
+//   32 f32 accumulator registers per lane.
+// - split-K: each workgroup reduces its K-slice and atomicAdd's the
+//   32x256 fp32 partial (atomics are a rounding-free fp32 add; the
+//   slab is tiny so contention is negligible — guide §6 G12).
+// - x rows are assembled from the NHWC u8 frame tensor: one im2col
+//   row = 8 segments of 32 contiguous bytes; 256 threads stage a
+//   16-row chunk into LDS as bf16 (dequant scale fused into the
+//   u8->bf16 conversion).
+#include "common.h"
+
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
+#define CONV1_CIN 4
+#define CONV1_COUT 32
+#define CONV1_KSZ 8
+#define CONV1_STRIDE 4
+#define CONV1_OHW 20          // output spatial (84-8)/4+1
+#define CONV1_POS (CONV1_OHW * CONV1_OHW)       // 400 positions
+#define CONV1_N (CONV1_KSZ * CONV1_KSZ * CONV1_CIN)  // 256 patch size
+#define KC 16                 // k-rows staged per iteration
+
+__device__ __forceinline__ __bf16 u8_bf16(unsigned char v, float scale) {
+  return (__bf16)((float)v * scale);
+}
+
+// dy: [K, 32] bf16 row-major (K = B*400, NHWC conv output layout)
+// frames: [B, 84, 84, 4] u8 (NHWC)
+// out: [32, 256] fp32, PRE-ZEROED, atomicAdd target
+__global__ __launch_bounds__(256)
+void conv1_wrw_kernel(const __bf16* __restrict__ dy,
+                      const unsigned char* __restrict__ frames,
+                      float* __restrict__ out, int64_t K,
+                      int64_t k_per_wg, float scale) {
+  __shared__ __bf16 lds[KC * CONV1_COUT + KC * CONV1_N];  // dy | x
+  __bf16* s_dy = lds;                    // [KC][32]
+  __bf16* s_x = lds + KC * CONV1_COUT;   // [KC][256]
+
+  const int tid = threadIdx.x;
+  const int wave = tid / MA_WAVE;  // 0..3
+  const int lane = tid % MA_WAVE;
+  const int mt = wave & 1;         // M-tile (0..1): rows mt*16..mt*16+15
+  const int ng = wave >> 1;        // N-group (0..1): tiles ng*8..ng*8+7
+
+  f32x4 acc[8];
+#pragma unroll
+  for (int i = 0; i < 8; ++i) acc[i] = (f32x4)(0.0f);
+
+  const int64_t k_begin = (int64_t)blockIdx.x * k_per_wg;
+  const int64_t k_end = min(k_begin + k_per_wg, K);
+
+  for (int64_t kc = k_begin; kc < k_end; kc += KC) {
+    // ---- stage dy chunk: 16 rows x 32 bf16 (64B/row) --------------
+    // threads 0..63: each loads 16B (8 bf16)
+    if (tid < 64) {
+      int row = tid >> 2;        // 0..15
+      int seg = tid & 3;         // 0..3 (8 bf16 each)
+      int64_t kk = kc + row;
+      bf16x8 v = (bf16x8)(__bf16)0.0f;
+      if (kk < k_end) {
+        v = *(const bf16x8*)(dy + kk * CONV1_COUT + seg * 8);
+      }
+      *(bf16x8*)(s_dy + row * CONV1_COUT + seg * 8) = v;
+    }
+    // ---- stage x chunk: 16 rows x 256 bf16 from u8 patches --------
+    // each im2col row: 8 segments of 32 u8; 128 segments total;
+    // 256 threads -> each converts one 16-byte half segment
+    {
+      int seg = tid >> 1;        // 0..127
+      int half = tid & 1;        // 0..1
+      int row = seg >> 3;        // k-row 0..15
+      int r = seg & 7;           // patch row 0..7
+      int64_t kk = kc + row;
+      __bf16 vals[16];
+      if (kk < k_end) {
+        int64_t b = kk / CONV1_POS;
+        int pos = (int)(kk % CONV1_POS);
+        int oh = pos / CONV1_OHW, ow = pos % CONV1_OHW;
+        const unsigned char* src =
+            frames
+            + ((b * 84 + (int64_t)oh * CONV1_STRIDE + r) * 84
+               + (int64_t)ow * CONV1_STRIDE) * CONV1_CIN
+            + half * 16;
+        uint4 raw = *(const uint4*)src;
+        const unsigned char* bytes = (const unsigned char*)&raw;
+#pragma unroll
+        for (int j = 0; j < 16; ++j) vals[j] = u8_bf16(bytes[j], scale);
+      } else {
+#pragma unroll
+        for (int j = 0; j < 16; ++j) vals[j] = (__bf16)0.0f;
+      }
+      bf16x8* dst = (bf16x8*)(s_x + row * CONV1_N + r * 32 + half * 16);
+      dst[0] = *(bf16x8*)&vals[0];
+      dst[1] = *(bf16x8*)&vals[8];
+    }
+    __syncthreads();
+
+    // ---- MFMA: A[m,k]=dy[k, mt*16+m] (transposed read), B[k,n]=x --
+    // A fragment (16x16x32): lane holds m=lane%16, k=(lane/16)*8+j
+    bf16x8 a_frag;
+    {
+      int m = lane & 15;
+      int k0 = (lane >> 4) * 8;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        a_frag[j] = s_dy[(k0 + j) * CONV1_COUT + mt * 16 + m];
+      }
+    }
+#pragma unroll
+    for (int nt = 0; nt < 8; ++nt) {
+      int n0 = (ng * 8 + nt) * 16;
+      bf16x8 b_frag;
+      int n = lane & 15;
+      int k0 = (lane >> 4) * 8;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        b_frag[j] = s_x[(k0 + j) * CONV1_N + n0 + n];
+      }
+      acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+          a_frag, b_frag, acc[nt], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  // ---- epilogue: atomicAdd partials (C/D map: col=lane&15,
+  // row=(lane>>4)*4+reg) ------------------------------------------
+#pragma unroll
+  for (int nt = 0; nt < 8; ++nt) {
+    int col = (ng * 8 + nt) * 16 + (lane & 15);
+    int row_base = mt * 16 + (lane >> 4) * 4;
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) {
+      atomicAdd(&out[(row_base + reg) * CONV1_N + col], acc[nt][reg]);
+    }
+  }
+}
+
+// reorder [32][r*32+c*4+ci] fp32 -> conv weight grad [32][4][8][8]
+__global__ void conv1_wrw_reorder_kernel(const float* __restrict__ in,
+                                         float* __restrict__ out) {
+  int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= CONV1_COUT * CONV1_N) return;
+  int co = i / CONV1_N;
+  int rem = i % CONV1_N;
+  int r = rem / 32;
+  int c = (rem % 32) / CONV1_CIN;
+  int ci = rem % CONV1_CIN;
+  out[((co * CONV1_CIN + ci) * CONV1_KSZ + r) * CONV1_KSZ + c] =
+      in[i];
+}
+
+void conv1_wrw_launch(const void* dy, const unsigned char* frames,
+                      float* scratch, float* grad_w, int64_t K,
+                      float scale, hipStream_t stream) {
+  HIP_CHECK(hipMemsetAsync(scratch, 0,
+                           CONV1_COUT * CONV1_N * sizeof(float), stream));
+  // split-K: target ~2048 workgroups (8 XCDs x 32 CUs x 8 blocks)
+  int64_t k_per_wg = (K + 2047) / 2048;
+  k_per_wg = ((k_per_wg + KC - 1) / KC) * KC;
+  if (k_per_wg < KC) k_per_wg = KC;
+  int grid = (int)((K + k_per_wg - 1) / k_per_wg);
+  hipLaunchKernelGGL(conv1_wrw_kernel, dim3(grid), dim3(256), 0, stream,
+                     (const __bf16*)dy, frames, scratch, K, k_per_wg,
+                     scale);
+  hipLaunchKernelGGL(conv1_wrw_reorder_kernel,
+                     dim3((CONV1_COUT * CONV1_N + 255) / 256), dim3(256),
+                     0, stream, scratch, grad_w);
+}
+
+// ---------------------------------------------------------------------
+// layout probe: one mfma_f32_16x16x32_bf16 tile, D = A[16,32] @ B[32,16]
+// with the fragment maps assumed above. Used by tests to pin the
+// lane->element mapping before trusting the conv kernel.
+// ---------------------------------------------------------------------
+__global__ void mfma_probe_kernel(const __bf16* __restrict__ A,
+                                  const __bf16* __restrict__ B,
+                                  float* __restrict__ D) {
+  int lane = threadIdx.x;
+  bf16x8 a_frag, b_frag;
+  int m = lane & 15;
+  int k0 = (lane >> 4) * 8;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    a_frag[j] = A[m * 32 + k0 + j];       // A[m][k] row-major
+    b_frag[j] = B[(k0 + j) * 16 + (lane & 15)];  // B[k][n] row-major
+  }
+  f32x4 acc = (f32x4)(0.0f);
+  acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_frag, b_frag, acc, 0, 0,
+                                                0);
+  int col = lane & 15;
+  int row_base = (lane >> 4) * 4;
+#pragma unroll
+  for (int reg = 0; reg < 4; ++reg) {
+    D[(row_base + reg) * 16 + col] = acc[reg];
+  }
+}
+
+void mfma_probe_launch(const void* A, const void* B, float* D,
+                       hipStream_t stream) {
+  hipLaunchKernelGGL(mfma_probe_kernel, dim3(1), dim3(64), 0, stream,
+                     (const __bf16*)A, (const __bf16*)B, D);
+}

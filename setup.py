@@ -25,6 +25,7 @@ ext = CUDAExtension(
         os.path.join(HIP_DIR, "multi_tensor.hip"),
         os.path.join(HIP_DIR, "distributions.hip"),
         os.path.join(HIP_DIR, "elementwise.hip"),
+        os.path.join(HIP_DIR, "conv1_wrw.hip"),
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],

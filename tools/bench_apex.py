@@ -86,14 +86,17 @@ def main():
         sn = ops.dequant_u8(
             batch["next_state"].permute(0, 2, 3, 1)
         ).view(-1, 84, 84, 4).permute(0, 3, 1, 2)
-        with t.autocast(device_type="cuda", dtype=t.bfloat16):
-            with t.no_grad():
+        # no_grad OUTSIDE autocast: autocast's weight-cast cache must
+        # not leak grad-less casts into the training forward
+        with t.no_grad():
+            with t.autocast(device_type="cuda", dtype=t.bfloat16):
                 online_next = qnet(sn).float()
                 best = online_next.argmax(dim=1, keepdim=True)
                 q_next = qnet_t(sn).float().gather(1, best)
-                y = batch["reward"].view(-1, 1) + 0.99 * (
-                    1.0 - batch["terminal"].view(-1, 1)
-                ) * q_next
+            y = batch["reward"].view(-1, 1) + 0.99 * (
+                1.0 - batch["terminal"].view(-1, 1)
+            ) * q_next
+        with t.autocast(device_type="cuda", dtype=t.bfloat16):
             q = qnet(s).float().gather(1, batch["action"])
         td = q - y
         loss = (td.pow(2).view(-1) * is_w).mean()

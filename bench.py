@@ -43,8 +43,12 @@ class ImpalaLearnerBench:
         discount: float = 0.99,
         grad_clip: float = 40.0,
         capturable: bool = False,
+        fused_stem: bool = True,
     ):
-        from machin_amd.model.nets.nature_cnn import ActorCriticCNN
+        from machin_amd.model.nets.nature_cnn import (
+            ActorCriticCNN,
+            FusedActorCriticCNN,
+        )
         import machin_amd.ops as ops
 
         if not ops.available():
@@ -62,7 +66,11 @@ class ImpalaLearnerBench:
         self.discount = discount
         self.grad_clip = grad_clip
 
-        self.model = ActorCriticCNN(frames, action_num).to(self.device)
+        self.fused_stem = fused_stem and dtype == t.bfloat16 and frames == 4
+        if self.fused_stem:
+            self.model = FusedActorCriticCNN(action_num).to(self.device)
+        else:
+            self.model = ActorCriticCNN(frames, action_num).to(self.device)
         self.model = self.model.to(memory_format=t.channels_last)
         self.reducer = None
         if distributed:
@@ -123,19 +131,19 @@ class ImpalaLearnerBench:
     def _step_body(self, data):
         T, B = self.unroll, self.env_batch
 
-        if self.dtype == t.bfloat16:
-            # fused u8->bf16 dequant (gfx950 kernel): one vectorized
-            # pass instead of .to() + .mul_(). The pool frames are
-            # channels_last; permute exposes their linear memory so
-            # the kernel runs on the raw bytes and the result keeps
-            # the NHWC stride order MIOpen wants.
-            nhwc = data["frames"].permute(0, 2, 3, 1)
-            out = self.ops.dequant_u8(nhwc, 1.0 / 255.0)
-            frames = out.view(nhwc.shape).permute(0, 3, 1, 2)
+        if self.fused_stem:
+            # raw u8 frames straight into the fused stem (dequant and
+            # the stem conv weight gradient are hand-written kernels)
+            logits, values = self.model(data["frames"])
         else:
-            frames = data["frames"].to(self.dtype).mul_(1.0 / 255.0)
-        with t.autocast(device_type="cuda", dtype=self.dtype):
-            logits, values = self.model(frames)
+            if self.dtype == t.bfloat16:
+                nhwc = data["frames"].permute(0, 2, 3, 1)
+                out = self.ops.dequant_u8(nhwc, 1.0 / 255.0)
+                frames = out.view(nhwc.shape).permute(0, 3, 1, 2)
+            else:
+                frames = data["frames"].to(self.dtype).mul_(1.0 / 255.0)
+            with t.autocast(device_type="cuda", dtype=self.dtype):
+                logits, values = self.model(frames)
         logits = logits.float().view(T, B, -1)
         values = values.float().view(T, B)
         log_pi = t.log_softmax(logits, dim=-1)
@@ -207,6 +215,8 @@ def main():
     parser.add_argument("--actions", type=int, default=6)
     parser.add_argument("--graph", action="store_true",
                         help="capture the learner step in a hipGraph")
+    parser.add_argument("--no-fused-stem", action="store_true",
+                        help="disable the fused u8 stem conv")
     args = parser.parse_args()
 
     world_size = int(os.environ.get("WORLD_SIZE", "1"))
@@ -231,6 +241,7 @@ def main():
         action_num=args.actions,
         distributed=distributed,
         capturable=args.graph and not distributed,
+        fused_stem=not args.no_fused_stem,
     )
 
     if args.graph and not distributed:

@@ -66,3 +66,34 @@ def mlp(sizes: Sequence[int], activation=nn.ReLU, output_activation=None):
         elif output_activation is not None:
             layers.append(output_activation())
     return nn.Sequential(*layers)
+
+
+class FusedActorCriticCNN(NeuralNetworkModule):
+    """ActorCriticCNN variant consuming RAW uint8 frames: the stem is
+    machin_amd.ops.fused_conv.FusedAtariConv1 (fused u8 dequant +
+    conv, hand-written gfx950 weight-gradient kernel); the rest of the
+    stack is identical to ActorCriticCNN."""
+
+    def __init__(self, action_num: int = 6, feature_dim: int = 512):
+        super().__init__()
+        from ...ops.fused_conv import FusedAtariConv1
+
+        self.stem = FusedAtariConv1()
+        self.conv_rest = nn.Sequential(
+            nn.Conv2d(32, 64, 4, stride=2),
+            nn.ReLU(inplace=True),
+            nn.Conv2d(64, 64, 3, stride=1),
+            nn.ReLU(inplace=True),
+        )
+        self.fc = nn.Sequential(
+            nn.Flatten(), nn.Linear(64 * 7 * 7, feature_dim),
+            nn.ReLU(inplace=True),
+        )
+        self.policy = nn.Linear(feature_dim, action_num)
+        self.value = nn.Linear(feature_dim, 1)
+
+    def forward(self, frames_u8: t.Tensor):
+        x = t.relu(self.stem(frames_u8))
+        with t.autocast(device_type="cuda", dtype=t.bfloat16):
+            feat = self.fc(self.conv_rest(x))
+            return self.policy(feat), self.value(feat)

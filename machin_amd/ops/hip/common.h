@@ -2,6 +2,7 @@
 #pragma once
 #include <hip/hip_runtime.h>
 #include <cstdint>
+#include <cstdlib>
 
 #define MA_WAVE 64  // CDNA4 wavefront width (not 32)
 

@@ -32,7 +32,7 @@ typedef __attribute__((ext_vector_type(4))) float f32x4;
 #define CONV1_OHW 20          // output spatial (84-8)/4+1
 #define CONV1_POS (CONV1_OHW * CONV1_OHW)       // 400 positions
 #define CONV1_N (CONV1_KSZ * CONV1_KSZ * CONV1_CIN)  // 256 patch size
-#define KC 16                 // k-rows staged per iteration
+#define KC 32  // k-rows per chunk: one full MFMA K-step (16x16x32)
 
 __device__ __forceinline__ __bf16 u8_bf16(unsigned char v, float scale) {
   return (__bf16)((float)v * scale);
@@ -64,28 +64,35 @@ void conv1_wrw_kernel(const __bf16* __restrict__ dy,
   const int64_t k_end = min(k_begin + k_per_wg, K);
 
   for (int64_t kc = k_begin; kc < k_end; kc += KC) {
-    // ---- stage dy chunk: 16 rows x 32 bf16 (64B/row) --------------
-    // threads 0..63: each loads 16B (8 bf16)
-    if (tid < 64) {
-      int row = tid >> 2;        // 0..15
+    // ---- stage dy chunk: 32 rows x 32 bf16 (64B/row) --------------
+    // threads 0..127: each loads 16B (8 bf16)
+    if (tid < 128) {
+      int row = tid >> 2;        // 0..31
       int seg = tid & 3;         // 0..3 (8 bf16 each)
       int64_t kk = kc + row;
-      bf16x8 v = (bf16x8)(__bf16)0.0f;
+      bf16x8 v;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) v[j] = (__bf16)0.0f;
       if (kk < k_end) {
         v = *(const bf16x8*)(dy + kk * CONV1_COUT + seg * 8);
       }
       *(bf16x8*)(s_dy + row * CONV1_COUT + seg * 8) = v;
     }
-    // ---- stage x chunk: 16 rows x 256 bf16 from u8 patches --------
-    // each im2col row: 8 segments of 32 u8; 128 segments total;
-    // 256 threads -> each converts one 16-byte half segment
+    // ---- stage x chunk: 32 rows x 256 bf16 from u8 patches --------
+    // each im2col row: 8 segments of 32 u8; 256 segments total;
+    // 256 threads -> each converts one full 32-byte segment
     {
-      int seg = tid >> 1;        // 0..127
-      int half = tid & 1;        // 0..1
-      int row = seg >> 3;        // k-row 0..15
-      int r = seg & 7;           // patch row 0..7
+      int row = tid >> 3;        // k-row 0..31
+      int r = tid & 7;           // patch row 0..7
       int64_t kk = kc + row;
-      __bf16 vals[16];
+      bf16x8 q0, q1, q2, q3;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        q0[j] = (__bf16)0.0f;
+        q1[j] = (__bf16)0.0f;
+        q2[j] = (__bf16)0.0f;
+        q3[j] = (__bf16)0.0f;
+      }
       if (kk < k_end) {
         int64_t b = kk / CONV1_POS;
         int pos = (int)(kk % CONV1_POS);
@@ -93,19 +100,30 @@ void conv1_wrw_kernel(const __bf16* __restrict__ dy,
         const unsigned char* src =
             frames
             + ((b * 84 + (int64_t)oh * CONV1_STRIDE + r) * 84
-               + (int64_t)ow * CONV1_STRIDE) * CONV1_CIN
-            + half * 16;
-        uint4 raw = *(const uint4*)src;
-        const unsigned char* bytes = (const unsigned char*)&raw;
+               + (int64_t)ow * CONV1_STRIDE) * CONV1_CIN;
+        uint4 raw0 = *(const uint4*)src;
+        uint4 raw1 = *(const uint4*)(src + 16);
+        unsigned int words[8] = {raw0.x, raw0.y, raw0.z, raw0.w,
+                                 raw1.x, raw1.y, raw1.z, raw1.w};
 #pragma unroll
-        for (int j = 0; j < 16; ++j) vals[j] = u8_bf16(bytes[j], scale);
-      } else {
+        for (int w = 0; w < 2; ++w) {
 #pragma unroll
-        for (int j = 0; j < 16; ++j) vals[j] = (__bf16)0.0f;
+          for (int j = 0; j < 4; ++j) {
+            q0[w * 4 + j] = u8_bf16((words[w] >> (8 * j)) & 0xFF, scale);
+            q1[w * 4 + j] =
+                u8_bf16((words[w + 2] >> (8 * j)) & 0xFF, scale);
+            q2[w * 4 + j] =
+                u8_bf16((words[w + 4] >> (8 * j)) & 0xFF, scale);
+            q3[w * 4 + j] =
+                u8_bf16((words[w + 6] >> (8 * j)) & 0xFF, scale);
+          }
+        }
       }
-      bf16x8* dst = (bf16x8*)(s_x + row * CONV1_N + r * 32 + half * 16);
-      dst[0] = *(bf16x8*)&vals[0];
-      dst[1] = *(bf16x8*)&vals[8];
+      bf16x8* dst = (bf16x8*)(s_x + row * CONV1_N + r * 32);
+      dst[0] = q0;
+      dst[1] = q1;
+      dst[2] = q2;
+      dst[3] = q3;
     }
     __syncthreads();
 
@@ -169,16 +187,20 @@ void conv1_wrw_launch(const void* dy, const unsigned char* frames,
   HIP_CHECK(hipMemsetAsync(scratch, 0,
                            CONV1_COUT * CONV1_N * sizeof(float), stream));
   // split-K: target ~2048 workgroups (8 XCDs x 32 CUs x 8 blocks)
-  int64_t k_per_wg = (K + 2047) / 2048;
+  int64_t target_wg = 2048;
+  if (const char* e = getenv("MACHIN_CONV1_WG")) target_wg = atol(e);
+  int64_t k_per_wg = (K + target_wg - 1) / target_wg;
   k_per_wg = ((k_per_wg + KC - 1) / KC) * KC;
   if (k_per_wg < KC) k_per_wg = KC;
   int grid = (int)((K + k_per_wg - 1) / k_per_wg);
   hipLaunchKernelGGL(conv1_wrw_kernel, dim3(grid), dim3(256), 0, stream,
                      (const __bf16*)dy, frames, scratch, K, k_per_wg,
                      scale);
+  HIP_CHECK(hipGetLastError());
   hipLaunchKernelGGL(conv1_wrw_reorder_kernel,
                      dim3((CONV1_COUT * CONV1_N + 255) / 256), dim3(256),
                      0, stream, scratch, grad_w);
+  HIP_CHECK(hipGetLastError());
 }
 
 // ---------------------------------------------------------------------

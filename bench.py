@@ -43,7 +43,7 @@ class ImpalaLearnerBench:
         discount: float = 0.99,
         grad_clip: float = 40.0,
         capturable: bool = False,
-        fused_stem: bool = True,
+        fused_stem: bool = False,
     ):
         from machin_amd.model.nets.nature_cnn import (
             ActorCriticCNN,
@@ -215,8 +215,10 @@ def main():
     parser.add_argument("--actions", type=int, default=6)
     parser.add_argument("--graph", action="store_true",
                         help="capture the learner step in a hipGraph")
-    parser.add_argument("--no-fused-stem", action="store_true",
-                        help="disable the fused u8 stem conv")
+    parser.add_argument("--fused-stem", action="store_true",
+                        help="use the hand-written fused u8 stem conv "
+                             "(measured ~5%% behind MIOpen's tuned igemm "
+                             "end-to-end; kept for kernel-level testing)")
     args = parser.parse_args()
 
     world_size = int(os.environ.get("WORLD_SIZE", "1"))
@@ -241,7 +243,7 @@ def main():
         action_num=args.actions,
         distributed=distributed,
         capturable=args.graph and not distributed,
-        fused_stem=not args.no_fused_stem,
+        fused_stem=args.fused_stem,
     )
 
     if args.graph and not distributed:

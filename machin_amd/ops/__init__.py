@@ -64,12 +64,28 @@ def polyak_update_(
     """
     if not targets:
         return
-    if targets[0].is_cuda:
+    fused_t, fused_s, plain_t, plain_s = [], [], [], []
+    for tt, ss in zip(targets, sources):
+        # the HIP kernel walks flat storage: standard-contiguous
+        # fp32 pairs only (channels_last params etc. take _foreach)
+        if (
+            tt.is_cuda
+            and tt.is_contiguous()
+            and ss.is_contiguous()
+            and tt.dtype == t.float32
+            and ss.dtype == t.float32
+        ):
+            fused_t.append(tt)
+            fused_s.append(ss)
+        else:
+            plain_t.append(tt)
+            plain_s.append(ss)
+    if fused_t:
         ext = _require_ext()
-        ext.multi_tensor_polyak(targets, sources, float(tau))
-    else:
-        t._foreach_mul_(targets, 1.0 - tau)
-        t._foreach_add_(targets, sources, alpha=tau)
+        ext.multi_tensor_polyak(fused_t, fused_s, float(tau))
+    if plain_t:
+        t._foreach_mul_(plain_t, 1.0 - tau)
+        t._foreach_add_(plain_t, plain_s, alpha=tau)
 
 
 # ======================================================================

@@ -42,12 +42,12 @@ class _FusedConv1Fn(t.autograd.Function):
         # channels_last u8 frames: permute exposes the NHWC memory as
         # a standard-contiguous view (no copy)
         frames_nhwc = frames_u8.permute(0, 2, 3, 1).contiguous()
-        grad_w = ext.conv1_wrw(
+        grad_w, grad_b = ext.conv1_wrw(
             dy_rows.to(t.bfloat16).contiguous(), frames_nhwc, ctx.scale
-        ).to(weight.dtype)
-        grad_b = None
-        if ctx.has_bias:
-            grad_b = gy.float().sum(dim=(0, 2, 3))
+        )
+        grad_w = grad_w.to(weight.dtype)
+        if not ctx.has_bias:
+            grad_b = None
         return None, grad_w, grad_b, None
 
 

@@ -39,7 +39,7 @@ void ou_update_launch(float*, int64_t, float, float, float, float, uint64_t,
 void u8_to_bf16_scale_launch(const unsigned char*, void*, int64_t, float,
                              hipStream_t);
 void conv1_wrw_launch(const void*, const unsigned char*, float*, float*,
-                      int64_t, float, hipStream_t);
+                      float*, int64_t, float, hipStream_t);
 void mfma_probe_launch(const void*, const void*, float*, hipStream_t);
 
 namespace {
@@ -245,7 +245,7 @@ Tensor u8_to_bf16_scale(Tensor in, double scale) {
   return out;
 }
 
-Tensor conv1_wrw(Tensor dy, Tensor frames, double scale) {
+std::vector<Tensor> conv1_wrw(Tensor dy, Tensor frames, double scale) {
   // dy: [K, 32] bf16 (NHWC-flattened conv output grad);
   // frames: [B, 84, 84, 4] u8. Returns [32, 4, 8, 8] fp32 grad.
   TORCH_CHECK(dy.is_cuda() && dy.scalar_type() == at::kBFloat16 &&
@@ -261,10 +261,12 @@ Tensor conv1_wrw(Tensor dy, Tensor frames, double scale) {
                              dy.options().dtype(at::kFloat));
   Tensor grad_w = at::empty({32, 4, 8, 8},
                             dy.options().dtype(at::kFloat));
+  Tensor grad_b = at::empty({32}, dy.options().dtype(at::kFloat));
   conv1_wrw_launch(dy.data_ptr(), frames.data_ptr<unsigned char>(),
                    scratch.data_ptr<float>(), grad_w.data_ptr<float>(),
-                   dy.size(0), (float)scale, current_stream());
-  return grad_w;
+                   grad_b.data_ptr<float>(), dy.size(0), (float)scale,
+                   current_stream());
+  return {grad_w, grad_b};
 }
 
 Tensor mfma_probe(Tensor A, Tensor B) {

@@ -44,7 +44,8 @@ __device__ __forceinline__ __bf16 u8_bf16(unsigned char v, float scale) {
 __global__ __launch_bounds__(256)
 void conv1_wrw_kernel(const __bf16* __restrict__ dy,
                       const unsigned char* __restrict__ frames,
-                      float* __restrict__ out, int64_t K,
+                      float* __restrict__ out,
+                      float* __restrict__ b_out, int64_t K,
                       int64_t k_per_wg, float scale) {
   __shared__ __bf16 lds[KC * CONV1_COUT + KC * CONV1_N];  // dy | x
   __bf16* s_dy = lds;                    // [KC][32]
@@ -59,6 +60,7 @@ void conv1_wrw_kernel(const __bf16* __restrict__ dy,
   f32x4 acc[8];
 #pragma unroll
   for (int i = 0; i < 8; ++i) acc[i] = (f32x4)(0.0f);
+  float bias_acc = 0.0f;
 
   const int64_t k_begin = (int64_t)blockIdx.x * k_per_wg;
   const int64_t k_end = min(k_begin + k_per_wg, K);
@@ -127,6 +129,15 @@ void conv1_wrw_kernel(const __bf16* __restrict__ dy,
     }
     __syncthreads();
 
+    // ---- fused bias: wave 0 lanes 0..31 sum dy columns (dy is
+    // already in LDS; saves a separate 1 GB reduction pass) ---------
+    if (wave == 0 && lane < CONV1_COUT) {
+#pragma unroll
+      for (int k = 0; k < KC; ++k) {
+        bias_acc += (float)s_dy[k * CONV1_COUT + lane];
+      }
+    }
+
     // ---- MFMA: A[m,k]=dy[k, mt*16+m] (transposed read), B[k,n]=x --
     // A fragment (16x16x32): lane holds m=lane%16, k=(lane/16)*8+j
     bf16x8 a_frag;
@@ -156,6 +167,9 @@ void conv1_wrw_kernel(const __bf16* __restrict__ dy,
 
   // ---- epilogue: atomicAdd partials (C/D map: col=lane&15,
   // row=(lane>>4)*4+reg) ------------------------------------------
+  if (wave == 0 && lane < CONV1_COUT) {
+    atomicAdd(&b_out[lane], bias_acc);
+  }
 #pragma unroll
   for (int nt = 0; nt < 8; ++nt) {
     int col = (ng * 8 + nt) * 16 + (lane & 15);
@@ -182,10 +196,12 @@ __global__ void conv1_wrw_reorder_kernel(const float* __restrict__ in,
 }
 
 void conv1_wrw_launch(const void* dy, const unsigned char* frames,
-                      float* scratch, float* grad_w, int64_t K,
-                      float scale, hipStream_t stream) {
+                      float* scratch, float* grad_w, float* grad_b,
+                      int64_t K, float scale, hipStream_t stream) {
   HIP_CHECK(hipMemsetAsync(scratch, 0,
                            CONV1_COUT * CONV1_N * sizeof(float), stream));
+  HIP_CHECK(hipMemsetAsync(grad_b, 0, CONV1_COUT * sizeof(float),
+                           stream));
   // split-K: target ~2048 workgroups (8 XCDs x 32 CUs x 8 blocks)
   int64_t target_wg = 2048;
   if (const char* e = getenv("MACHIN_CONV1_WG")) target_wg = atol(e);
@@ -194,8 +210,8 @@ void conv1_wrw_launch(const void* dy, const unsigned char* frames,
   if (k_per_wg < KC) k_per_wg = KC;
   int grid = (int)((K + k_per_wg - 1) / k_per_wg);
   hipLaunchKernelGGL(conv1_wrw_kernel, dim3(grid), dim3(256), 0, stream,
-                     (const __bf16*)dy, frames, scratch, K, k_per_wg,
-                     scale);
+                     (const __bf16*)dy, frames, scratch, grad_b, K,
+                     k_per_wg, scale);
   HIP_CHECK(hipGetLastError());
   hipLaunchKernelGGL(conv1_wrw_reorder_kernel,
                      dim3((CONV1_COUT * CONV1_N + 255) / 256), dim3(256),

@@ -23,11 +23,13 @@ class TestConv1Wrw:
                            device=DEV)
         gy = (t.randn(batch, 20, 20, 32, device=DEV) * 0.1).to(t.bfloat16)
 
-        grad_w = ext.conv1_wrw(
+        grad_w, grad_b = ext.conv1_wrw(
             gy.reshape(-1, 32).contiguous(), frames.contiguous(),
             1.0 / 255.0,
         )
         assert grad_w.shape == (32, 4, 8, 8)
+        ref_b = gy.float().sum(dim=(0, 1, 2))
+        assert t.allclose(grad_b, ref_b, rtol=1e-2, atol=1e-2)
 
         # fp32 reference: same math, plain torch autograd
         x = (frames.permute(0, 3, 1, 2).float() / 255.0).requires_grad_(

@@ -113,14 +113,20 @@ def _dist_worker(rank: int, world_size: int, port: int, config_data: dict):
     os.environ["MASTER_ADDR"] = "127.0.0.1"
     os.environ["MASTER_PORT"] = str(port)
     t.set_num_threads(2)
+    import torch.distributed as dist
+
     from ..parallel.distributed.world import World
     from .envs.classic_control import make_dataset
 
     World(world_size=world_size, rank=rank, name=str(rank),
-          dist_backend="nccl" if t.cuda.is_available() else "gloo")
+          dist_backend="nccl" if t.cuda.is_available() else "gloo",
+          dist_timeout=1800.0)
     config = Config(**config_data)
     launcher = DistributedLauncher(config, dataset_factory=make_dataset)
     launcher.fit()
+    # keep this process's control-plane services alive until every
+    # member finishes (A3C/APEX peers keep calling them)
+    dist.barrier()
 
 
 def launch_distributed(config):

@@ -246,3 +246,86 @@ class TestStaleUpdateRejected:
 
     # NOTE: fresh entries accept updates — covered in
     # TestDistributedPrioritizedBuffer above.
+
+
+class TestRNNDistributedBuffers:
+    def test_rnn_distributed_windows(self):
+        def fn(rank, world):
+            from machin_amd.frame.buffers.rnn_buffers import (
+                RNNDistributedBuffer,
+            )
+
+            group = world.create_rpc_group("g", ["0", "1", "2"])
+            buf = RNNDistributedBuffer(
+                sample_length=3, buffer_name="rnnbuf", group=group,
+                buffer_size=50,
+            )
+            group.barrier()
+            episode = [
+                {
+                    "state": {"state": t.full((1, 4), float(i))},
+                    "action": {"action": t.zeros(1, 1)},
+                    "next_state": {"state": t.zeros(1, 4)},
+                    "reward": float(i),
+                    "terminal": i == 7,
+                }
+                for i in range(8)
+            ]
+            buf.store_episode(episode)
+            group.barrier()
+            out = None
+            if rank == 0:
+                bs, batch = buf.sample_batch(
+                    6, sample_attrs=["state", "reward"]
+                )
+                state = batch[0]["state"]
+                # [windows, length, dim]; consecutive steps in windows
+                assert state.dim() == 3 and state.shape[1] == 3
+                diffs = state[:, 1:, 0] - state[:, :-1, 0]
+                assert t.allclose(diffs, t.ones_like(diffs))
+                out = True
+            group.barrier()
+            return out
+
+        results = run_multi(fn)
+        assert results[0] is True
+
+    def test_rnn_distributed_prioritized(self):
+        def fn(rank, world):
+            from machin_amd.frame.buffers.rnn_buffers import (
+                RNNDistributedPrioritizedBuffer,
+            )
+
+            group = world.create_rpc_group("g", ["0", "1", "2"])
+            buf = RNNDistributedPrioritizedBuffer(
+                sample_length=3, buffer_name="rnnpbuf", group=group,
+                buffer_size=50,
+            )
+            group.barrier()
+            episode = [
+                {
+                    "state": {"state": t.full((1, 4), float(i))},
+                    "action": {"action": t.zeros(1, 1)},
+                    "next_state": {"state": t.zeros(1, 4)},
+                    "reward": 0.0,
+                    "terminal": i == 7,
+                }
+                for i in range(8)
+            ]
+            buf.store_episode(episode, priorities=np.ones(8))
+            group.barrier()
+            out = None
+            if rank == 0:
+                bs, batch, index, is_weight = buf.sample_batch(
+                    6, sample_attrs=["state"]
+                )
+                assert bs > 0
+                state = batch[0]["state"]
+                assert state.dim() == 3 and state.shape[1] == 3
+                buf.update_priority(np.full(bs, 2.0), index)
+                out = True
+            group.barrier()
+            return out
+
+        results = run_multi(fn)
+        assert results[0] is True

@@ -328,6 +328,8 @@ class World:
         if op == "lut_has":
             with self._lut_lock:
                 return request["key"] in self._lut
+        if op == "ping":
+            return True
         if op == "barrier_enter":
             return self._barrier_enter(
                 request["key"], request["count"],
@@ -379,6 +381,22 @@ class World:
 
     def get_rpc_group(self, group_name: str, target: str = None):
         return self.groups.get(group_name)
+
+    def check_peers(self, timeout: float = 5.0) -> Dict[str, bool]:
+        """Liveness probe: ping every member's control server.
+        (the reference has no heartbeat — TODO at
+        _world.py:594; this is the machin_amd equivalent.)"""
+        out = {}
+        for name in self.get_members():
+            if name == self.name:
+                out[name] = True
+                continue
+            try:
+                self._control(name, {"op": "ping"}, timeout)
+                out[name] = True
+            except Exception:  # noqa: BLE001 - liveness probe
+                out[name] = False
+        return out
 
     def get_ranks(self) -> List[int]:
         return list(range(self.world_size))

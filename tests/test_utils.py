@@ -207,3 +207,17 @@ class TestPoolRespawn:
             assert pool.apply(lambda: 42, ()) == 42
         finally:
             pool.terminate()
+
+
+class TestTracing:
+    def test_noop_on_cpu(self):
+        from machin_amd.utils.tracing import trace_range, traced
+
+        with trace_range("phase"):
+            x = 1 + 1
+
+        @traced("fn")
+        def f(a):
+            return a * 2
+
+        assert f(21) == 42

@@ -223,3 +223,12 @@ class TestWorldBasics:
         results = run_multi(fn, names=["alpha", "beta", "gamma"])
         assert results[0][0] == ["alpha", "beta", "gamma"]
         assert results[1][1] == "beta"
+
+
+class TestPeerLiveness:
+    def test_check_peers(self):
+        def fn(rank, world):
+            status = world.check_peers()
+            return all(status.values()) and len(status) == 3
+
+        assert all(run_multi(fn))

@@ -1,0 +1,101 @@
+"""Shared-memory rollout ring: host actor farm -> GPU learner staging.
+
+SURVEY.md §2.6-C / §5.8: the reference ships rollouts as pickled CPU
+tensors over RPC (machin/frame/buffers/buffer_d.py:194-197). On
+MI355X the actor->learner path is: actor processes write rollout
+segments into preallocated SHARED-MEMORY slots (zero serialization),
+only slot indices cross process boundaries (tiny pipe messages), and
+the learner copies ready slots into pinned staging and issues ONE
+async H2D copy per attribute onto a side stream.
+
+Layout: for each attribute, one shared tensor of shape
+``[slots, *per_slot_shape]``. Actors acquire a free slot, fill it,
+mark it ready; the learner drains ready slots in batches and recycles
+them.
+"""
+from typing import Dict, List, Tuple
+
+import torch as t
+import torch.multiprocessing as mp
+
+
+class RolloutRing:
+    """Multi-producer single-consumer ring of rollout segments."""
+
+    def __init__(
+        self,
+        slots: int,
+        spec: Dict[str, Tuple[Tuple[int, ...], t.dtype]],
+        ctx=None,
+    ):
+        ctx = ctx or mp.get_context("spawn")
+        self.slots = int(slots)
+        self.spec = dict(spec)
+        self.data = {
+            k: t.zeros((self.slots, *shape), dtype=dtype).share_memory_()
+            for k, (shape, dtype) in self.spec.items()
+        }
+        self._free = ctx.Queue()
+        self._ready = ctx.Queue()
+        for i in range(self.slots):
+            self._free.put(i)
+
+    # -- actor side ----------------------------------------------------
+    def acquire(self, timeout: float = None) -> int:
+        """Take a free slot index (blocks when the learner lags)."""
+        return self._free.get(timeout=timeout)
+
+    def slot(self, index: int) -> Dict[str, t.Tensor]:
+        """Views of one slot's attribute tensors (write in place)."""
+        return {k: v[index] for k, v in self.data.items()}
+
+    def commit(self, index: int):
+        """Mark a filled slot ready for the learner."""
+        self._ready.put(index)
+
+    # -- learner side --------------------------------------------------
+    def drain(self, max_slots: int, timeout: float = 1.0) -> List[int]:
+        """Collect up to ``max_slots`` ready slot indices; waits for at
+        least one up to ``timeout``."""
+        out = []
+        try:
+            out.append(self._ready.get(timeout=timeout))
+        except Exception:  # noqa: BLE001 - queue.Empty
+            return out
+        while len(out) < max_slots:
+            try:
+                out.append(self._ready.get_nowait())
+            except Exception:  # noqa: BLE001
+                break
+        return out
+
+    def gather(
+        self,
+        indices: List[int],
+        device: t.device,
+        pinned: Dict[str, t.Tensor] = None,
+        non_blocking: bool = True,
+    ) -> Dict[str, t.Tensor]:
+        """Copy the chosen slots to ``device`` (through a pinned
+        staging buffer when provided) and return the batch."""
+        idx = t.tensor(indices, dtype=t.long)
+        out = {}
+        for k, buf in self.data.items():
+            host = buf.index_select(0, idx)
+            if pinned is not None and k in pinned:
+                staging = pinned[k][: len(indices)]
+                staging.copy_(host)
+                host = staging
+            out[k] = host.to(device, non_blocking=non_blocking)
+        return out
+
+    def release(self, indices: List[int]):
+        """Recycle consumed slots back to the actors."""
+        for i in indices:
+            self._free.put(i)
+
+    def make_pinned_staging(self, max_batch: int) -> Dict[str, t.Tensor]:
+        return {
+            k: t.empty((max_batch, *shape), dtype=dtype).pin_memory()
+            for k, (shape, dtype) in self.spec.items()
+        }

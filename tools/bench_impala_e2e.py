@@ -31,27 +31,45 @@ SPEC = None  # built in main (needs torch types)
 
 def actor_loop(ring, actor_id: int, stop_flag, steps_counter):
     """Synthetic Atari actor: fills rollout segments in shared memory.
-    (A real actor would run ALE + the policy net; frame generation is
-    the stand-in for simulation work — no network access for ROMs.)"""
+    A real actor would run ALE + the policy net (no ROMs without
+    network access); here a pre-generated segment library is cycled
+    with cheap mutation, so the measured cost is the TRANSPORT
+    pipeline (shared-memory writes, slot bookkeeping), not RNG."""
+    import os
+
+    t.set_num_threads(1)
+    try:
+        os.nice(5)  # the learner's host thread has priority
+    except OSError:
+        pass
     t.manual_seed(actor_id)
     gen = t.Generator().manual_seed(actor_id)
+    library = [
+        {
+            "frames": t.randint(0, 256, (T_UNROLL, 4, 84, 84),
+                                dtype=t.uint8, generator=gen),
+            "actions": t.randint(0, 6, (T_UNROLL,), generator=gen),
+            "behavior_logp": -t.rand(T_UNROLL, generator=gen) * 2.0,
+            "rewards": t.rand(T_UNROLL, generator=gen),
+            "terminals": t.zeros(T_UNROLL),
+        }
+        for _ in range(8)
+    ]
+    i = 0
     while not stop_flag[0]:
         try:
             slot_id = ring.acquire(timeout=1.0)
         except Exception:  # noqa: BLE001 - queue.Empty on shutdown
             continue
+        src = library[i % len(library)]
+        i += 1
         slot = ring.slot(slot_id)
-        t.randint(
-            0, 256, slot["frames"].shape, dtype=t.uint8, generator=gen,
-            out=slot["frames"],
-        )
-        t.randint(
-            0, 6, slot["actions"].shape, generator=gen,
-            out=slot["actions"],
-        )
-        slot["behavior_logp"].uniform_(-2.0, 0.0, generator=gen)
-        slot["rewards"].uniform_(0.0, 1.0, generator=gen)
-        slot["terminals"].zero_()
+        slot["frames"].copy_(src["frames"])
+        slot["frames"][0, 0, 0, 0] = i % 256  # cheap per-segment variation
+        slot["actions"].copy_(src["actions"])
+        slot["behavior_logp"].copy_(src["behavior_logp"])
+        slot["rewards"].copy_(src["rewards"])
+        slot["terminals"].copy_(src["terminals"])
         ring.commit(slot_id)
         with steps_counter.get_lock():
             steps_counter.value += T_UNROLL
@@ -59,7 +77,7 @@ def actor_loop(ring, actor_id: int, stop_flag, steps_counter):
 
 def main():
     parser = argparse.ArgumentParser()
-    parser.add_argument("--actors", type=int, default=12)
+    parser.add_argument("--actors", type=int, default=8)
     parser.add_argument("--seconds", type=float, default=20.0)
     parser.add_argument("--env-batch", type=int, default=1024)
     parser.add_argument("--pool-segments", type=int, default=4096)

@@ -45,4 +45,8 @@ def __getattr__(name):
         from .assigner import ModelAssigner
 
         return ModelAssigner
+    if name == "RolloutRing":
+        from .rollout_ring import RolloutRing
+
+        return RolloutRing
     raise AttributeError(name)

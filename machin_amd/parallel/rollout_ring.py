@@ -99,3 +99,34 @@ class RolloutRing:
             k: t.empty((max_batch, *shape), dtype=dtype).pin_memory()
             for k, (shape, dtype) in self.spec.items()
         }
+
+    def host_register(self) -> bool:
+        """Learner-side: hipHostRegister the shared-memory buffers so
+        H2D copies DMA STRAIGHT from the ring slots (no host staging
+        copy at all). Returns False when registration is unavailable
+        (callers then use pinned staging via :meth:`gather`)."""
+        try:
+            cudart = t.cuda.cudart()
+            for buf in self.data.values():
+                rc = cudart.cudaHostRegister(
+                    buf.data_ptr(), buf.numel() * buf.element_size(), 0
+                )
+                if int(rc) != 0:
+                    return False
+            return True
+        except Exception:  # noqa: BLE001 - optional fast path
+            return False
+
+    def upload_slots(
+        self,
+        indices: List[int],
+        pool: Dict[str, t.Tensor],
+        positions: List[int],
+        non_blocking: bool = True,
+    ):
+        """Direct DMA: copy each ready slot into its HBM pool position
+        (requires :meth:`host_register`; issue on a side stream)."""
+        for slot_id, pos in zip(indices, positions):
+            for k, buf in self.data.items():
+                pool[k][pos].copy_(buf[slot_id],
+                                   non_blocking=non_blocking)

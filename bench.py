@@ -88,7 +88,10 @@ class ImpalaLearnerBench:
         # synthetic rollout pool (uint8 frames like a real Atari actor
         # feed; behavior log-probs from a slightly-off policy)
         TB = unroll * env_batch
-        g = t.Generator(device="cpu").manual_seed(1234)
+        # per-rank data seeds: ranks must NOT train on identical data,
+        # or the all-reduce degenerates to a no-op check
+        rank = int(os.environ.get("RANK", "0"))
+        g = t.Generator(device="cpu").manual_seed(1234 + rank * 1000)
         self.pool = []
         for _ in range(pool_size):
             self.pool.append(

@@ -1,0 +1,3 @@
+from . import envs, utils, wrappers
+
+__all__ = ["envs", "utils", "wrappers"]

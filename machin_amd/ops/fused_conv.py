@@ -23,11 +23,22 @@ class _FusedConv1Fn(t.autograd.Function):
     def forward(ctx, frames_u8: t.Tensor, weight: t.Tensor,
                 bias: t.Tensor, scale: float):
         # frames_u8: [B, 4, 84, 84] channels_last uint8
-        nhwc = frames_u8.permute(0, 2, 3, 1)  # contiguous view
-        x = dequant_u8(nhwc, scale).view(nhwc.shape).permute(0, 3, 1, 2)
-        y = F.conv2d(x, weight.to(t.bfloat16),
-                     bias.to(t.bfloat16) if bias is not None else None,
-                     stride=4)
+        ext = _require_ext()
+        B = frames_u8.shape[0]
+        nhwc = frames_u8.permute(0, 2, 3, 1).contiguous()
+        # patch-major weight repack: n = r*32 + c*4 + ci
+        w_rs = (
+            weight.permute(2, 3, 1, 0).reshape(256, 32)
+            .to(t.bfloat16).contiguous()
+        )
+        y_rows = ext.conv1_fwd(
+            nhwc, w_rs,
+            bias.float().contiguous() if bias is not None
+            else t.empty(0),
+            scale,
+        )
+        # [K,32] IS the NHWC image; permute -> channels_last NCHW
+        y = y_rows.view(B, 20, 20, 32).permute(0, 3, 1, 2)
         ctx.save_for_backward(frames_u8, weight)
         ctx.scale = scale
         ctx.has_bias = bias is not None

@@ -41,6 +41,8 @@ void u8_to_bf16_scale_launch(const unsigned char*, void*, int64_t, float,
 void conv1_wrw_launch(const void*, const unsigned char*, float*, float*,
                       float*, int64_t, float, hipStream_t);
 void mfma_probe_launch(const void*, const void*, float*, hipStream_t);
+void conv1_fwd_launch(const unsigned char*, const void*, const float*,
+                      void*, int64_t, float, hipStream_t);
 
 namespace {
 
@@ -269,6 +271,33 @@ std::vector<Tensor> conv1_wrw(Tensor dy, Tensor frames, double scale) {
   return {grad_w, grad_b};
 }
 
+Tensor conv1_fwd(Tensor frames, Tensor weight, Tensor bias,
+                 double scale) {
+  // frames: [B, 84, 84, 4] u8 NHWC; weight: [256, 32] bf16 (patch-major
+  // repack of the conv weight); bias: [32] fp32 or empty.
+  TORCH_CHECK(frames.is_cuda() && frames.scalar_type() == at::kByte &&
+                  frames.is_contiguous(),
+              "frames must be contiguous u8 CUDA");
+  TORCH_CHECK(weight.scalar_type() == at::kBFloat16 &&
+                  weight.is_contiguous() && weight.size(0) == 256 &&
+                  weight.size(1) == 32,
+              "weight must be contiguous [256,32] bf16");
+  const at::cuda::OptionalCUDAGuard guard(frames.device());
+  int64_t K = frames.size(0) * 400;
+  Tensor out = at::empty({K, 32},
+                         weight.options().dtype(at::kBFloat16));
+  const float* bias_ptr = nullptr;
+  if (bias.defined() && bias.numel() == 32) {
+    TORCH_CHECK(bias.scalar_type() == at::kFloat && bias.is_contiguous(),
+                "bias must be contiguous fp32");
+    bias_ptr = bias.data_ptr<float>();
+  }
+  conv1_fwd_launch(frames.data_ptr<unsigned char>(), weight.data_ptr(),
+                   bias_ptr, out.data_ptr(), K, (float)scale,
+                   current_stream());
+  return out;
+}
+
 Tensor mfma_probe(Tensor A, Tensor B) {
   TORCH_CHECK(A.is_cuda() && A.scalar_type() == at::kBFloat16 &&
               A.is_contiguous() && A.size(0) == 16 && A.size(1) == 32,
@@ -301,4 +330,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("u8_to_bf16_scale", &u8_to_bf16_scale);
   m.def("conv1_wrw", &conv1_wrw);
   m.def("mfma_probe", &mfma_probe);
+  m.def("conv1_fwd", &conv1_fwd);
 }

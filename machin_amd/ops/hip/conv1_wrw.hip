@@ -252,3 +252,119 @@ void mfma_probe_launch(const void* A, const void* B, float* D,
   hipLaunchKernelGGL(mfma_probe_kernel, dim3(1), dim3(64), 0, stream,
                      (const __bf16*)A, (const __bf16*)B, D);
 }
+
+// ---------------------------------------------------------------------
+// fused forward for the same stem conv: out[K,32] = im2col(x)[K,256]
+// @ W[256,32] + bias, x read directly as u8 (dequant fused).
+// Per workgroup: 64 output rows x all 32 channels; K-loop over the 8
+// patch rows (32 patch values each, one contiguous 32-byte u8
+// segment per (k-row, patch-row) — same addressing as the wrw
+// kernel). W (256x32 bf16, 16 KB) is staged to LDS once.
+// ---------------------------------------------------------------------
+#define FWD_ROWS 64  // k-rows (output pixels) per workgroup
+
+__global__ __launch_bounds__(256)
+void conv1_fwd_kernel(const unsigned char* __restrict__ frames,
+                      const __bf16* __restrict__ weight,  // [256][32]
+                      const float* __restrict__ bias,     // [32]
+                      __bf16* __restrict__ out,           // [K][32]
+                      int64_t K, float scale) {
+  __shared__ __bf16 s_x[FWD_ROWS * 32];   // one patch-row chunk
+  __shared__ __bf16 s_w[CONV1_N * 32];    // full weight, staged once
+
+  const int tid = threadIdx.x;
+  const int wave = tid / MA_WAVE;  // 0..3
+  const int lane = tid % MA_WAVE;
+  // wave -> (2 M-tiles of 16 rows) x (2 N-tiles of 16 cols): wave w
+  // owns M-tile pair row (w&1) and N-tile (w>>1)
+  const int wm = wave & 1;   // 0..1 -> rows wm*32 .. wm*32+31 (2 tiles)
+  const int wn = wave >> 1;  // 0..1 -> cols wn*16 .. wn*16+15
+
+  // stage the whole weight [256][32]
+  for (int i = tid; i < CONV1_N * 32 / 8; i += 256) {
+    ((bf16x8*)s_w)[i] = ((const bf16x8*)weight)[i];
+  }
+
+  const int64_t row0 = (int64_t)blockIdx.x * FWD_ROWS;
+  f32x4 acc[2];  // two M-tiles (16x16 each) per wave
+  acc[0] = (f32x4)(0.0f);
+  acc[1] = (f32x4)(0.0f);
+  __syncthreads();
+
+  for (int pr = 0; pr < CONV1_KSZ; ++pr) {  // patch rows = K chunks
+    // stage 64 k-rows x 32 patch values (u8 segment -> bf16)
+    {
+      int krow = tid >> 2;          // 0..63
+      int quarter = tid & 3;        // 8 values each
+      int64_t kk = row0 + krow;
+      bf16x8 q;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) q[j] = (__bf16)0.0f;
+      if (kk < K) {
+        int64_t b = kk / CONV1_POS;
+        int pos = (int)(kk % CONV1_POS);
+        int oh = pos / CONV1_OHW, ow = pos % CONV1_OHW;
+        const unsigned char* src =
+            frames
+            + ((b * 84 + (int64_t)oh * CONV1_STRIDE + pr) * 84
+               + (int64_t)ow * CONV1_STRIDE) * CONV1_CIN
+            + quarter * 8;
+        uint2 raw = *(const uint2*)src;
+        unsigned int words[2] = {raw.x, raw.y};
+#pragma unroll
+        for (int w = 0; w < 2; ++w) {
+#pragma unroll
+          for (int j = 0; j < 4; ++j) {
+            q[w * 4 + j] = u8_bf16((words[w] >> (8 * j)) & 0xFF, scale);
+          }
+        }
+      }
+      *(bf16x8*)(s_x + krow * 32 + quarter * 8) = q;
+    }
+    __syncthreads();
+
+    // MFMA over this chunk's K=32 (patch cols x channels)
+#pragma unroll
+    for (int mt = 0; mt < 2; ++mt) {
+      int m = wm * 32 + mt * 16 + (lane & 15);
+      int k0 = (lane >> 4) * 8;
+      // A[m][k]: contiguous along k -> one vector read
+      bf16x8 a_frag = *(bf16x8*)(s_x + m * 32 + k0);
+      bf16x8 b_frag;
+      int n = wn * 16 + (lane & 15);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        b_frag[j] = s_w[(pr * 32 + k0 + j) * 32 + n];
+      }
+      acc[mt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+          a_frag, b_frag, acc[mt], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  // epilogue: bias add, bf16 store (C/D map: col=lane&15,
+  // row=(lane>>4)*4+reg)
+#pragma unroll
+  for (int mt = 0; mt < 2; ++mt) {
+    int col = wn * 16 + (lane & 15);
+    float b = bias != nullptr ? bias[col] : 0.0f;
+    int row_base = wm * 32 + mt * 16 + (lane >> 4) * 4;
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) {
+      int64_t kk = row0 + row_base + reg;
+      if (kk < K) {
+        out[kk * 32 + col] = (__bf16)(acc[mt][reg] + b);
+      }
+    }
+  }
+}
+
+void conv1_fwd_launch(const unsigned char* frames, const void* weight,
+                      const float* bias, void* out, int64_t K,
+                      float scale, hipStream_t stream) {
+  int64_t grid = (K + FWD_ROWS - 1) / FWD_ROWS;
+  hipLaunchKernelGGL(conv1_fwd_kernel, dim3((unsigned)grid), dim3(256), 0,
+                     stream, frames, (const __bf16*)weight, bias,
+                     (__bf16*)out, K, scale);
+  HIP_CHECK(hipGetLastError());
+}

@@ -95,3 +95,27 @@ class TestMfmaProbe:
         assert t.allclose(D, ref, rtol=2e-2, atol=1e-2), (
             (D - ref).abs().max().item()
         )
+
+
+class TestConv1Fwd:
+    @pytest.mark.parametrize("batch", [1, 3, 64])
+    def test_vs_fp32_reference(self, batch):
+        from machin_amd.ops import _require_ext
+
+        ext = _require_ext()
+        t.manual_seed(5)
+        frames = t.randint(0, 256, (batch, 84, 84, 4), dtype=t.uint8,
+                           device=DEV)
+        w = t.randn(32, 4, 8, 8, device=DEV) * 0.05
+        b = t.randn(32, device=DEV) * 0.1
+        w_rs = w.permute(2, 3, 1, 0).reshape(256, 32).to(
+            t.bfloat16
+        ).contiguous()
+        y_rows = ext.conv1_fwd(frames.contiguous(), w_rs,
+                               b.contiguous(), 1.0 / 255.0)
+        y = y_rows.view(batch, 20, 20, 32).permute(0, 3, 1, 2).float()
+        x = frames.permute(0, 3, 1, 2).float() / 255.0
+        ref = F.conv2d(x, w, b, stride=4)
+        denom = ref.abs().max().clamp_min(1e-3)
+        rel = (y - ref).abs().max() / denom
+        assert rel.item() < 2e-2, f"rel err {rel.item()}"

@@ -353,3 +353,117 @@ class TestARSFullTrain:
 
         results = run_multi(fn, timeout=900)
         assert any(results), "ARS did not solve CartPole"
+
+
+class TestDDPGApexFullTrain:
+    def test_full_train(self):
+        """APEX-DDPG solves Pendulum (reference gate:
+        test/frame/algorithms/test_apex.py:349-358, smoothed > -400)."""
+        def fn(rank, world):
+            import time
+
+            import torch as t
+            import torch.nn as nn
+
+            from machin_amd.env.envs.classic_control import PendulumEnv
+            from machin_amd.frame.algorithms import DDPGApex
+            from machin_amd.frame.helpers.servers import model_server_helper
+
+            class Actor(nn.Module):
+                def __init__(self):
+                    super().__init__()
+                    self.fc1 = nn.Linear(3, 16)
+                    self.fc2 = nn.Linear(16, 16)
+                    self.fc3 = nn.Linear(16, 1)
+
+                def forward(self, state):
+                    a = t.relu(self.fc1(state))
+                    a = t.relu(self.fc2(a))
+                    return t.tanh(self.fc3(a)) * 2.0
+
+            class Critic(nn.Module):
+                def __init__(self):
+                    super().__init__()
+                    self.fc1 = nn.Linear(4, 16)
+                    self.fc2 = nn.Linear(16, 16)
+                    self.fc3 = nn.Linear(16, 1)
+
+                def forward(self, state, action):
+                    x = t.cat([state, action], dim=1)
+                    return self.fc3(t.relu(self.fc2(t.relu(self.fc1(x)))))
+
+            t.manual_seed(rank)
+            servers = model_server_helper(model_num=1)
+            group = world.create_rpc_group("apex", ["0", "1", "2"])
+            apex = DDPGApex(
+                Actor(), Actor(), Critic(), Critic(), t.optim.Adam,
+                nn.MSELoss(reduction="sum"), group, servers,
+                batch_size=100, update_rate=0.005,
+                actor_learning_rate=5e-4, critic_learning_rate=1e-3,
+                replay_size=20000,
+            )
+            group.barrier()
+            if rank == 2:
+                apex.set_sync(False)
+                deadline = time.monotonic() + 480
+                while (
+                    not group.is_paired("solved")
+                    and time.monotonic() < deadline
+                ):
+                    if apex.replay_buffer.all_size() > 500:
+                        apex.update()
+                    else:
+                        time.sleep(0.05)
+                solved = group.is_paired("solved")
+            else:
+                apex.set_sync(False)
+                env = PendulumEnv(seed=rank)
+                smoothed, wins = -1600.0, 0
+                solved = False
+                deadline = time.monotonic() + 480
+                while time.monotonic() < deadline:
+                    if group.is_paired("solved"):
+                        solved = True
+                        break
+                    obs = t.tensor(env.reset(), dtype=t.float32).view(1, 3)
+                    total = 0.0
+                    transitions = []
+                    done = False
+                    while not done:
+                        with t.no_grad():
+                            action = apex.act_with_noise(
+                                {"state": obs}, noise_param=(0.0, 0.3),
+                                mode="normal",
+                            ).clamp(-2, 2)
+                        o, r, done, _ = env.step(action.view(-1).numpy())
+                        o = t.tensor(o, dtype=t.float32).view(1, 3)
+                        total += r
+                        transitions.append(
+                            {
+                                "state": {"state": obs},
+                                "action": {"action": action.view(1, 1)},
+                                "next_state": {"state": o},
+                                "reward": r / 10.0,
+                                "terminal": False,
+                            }
+                        )
+                        obs = o
+                    apex.store_episode(transitions)
+                    apex.manual_sync()
+                    smoothed = smoothed * 0.9 + total * 0.1
+                    if smoothed > -400:
+                        wins += 1
+                        if wins >= 5:
+                            try:
+                                group.pair("solved", True)
+                            except RuntimeError:
+                                pass
+                            solved = True
+                            break
+                    else:
+                        wins = 0
+            group.barrier()
+            return solved
+
+        results = run_multi(fn, timeout=600)
+        assert any(results[:2]), "APEX-DDPG samplers never reached -400"

@@ -200,7 +200,9 @@ class DQN(TorchFramework):
             device = target.device
             reward = reward.to(device).float().view(batch_size, 1)
             terminal = terminal.to(device).float().view(batch_size, 1)
-            y = reward + self.discount * (1.0 - terminal) * target
+            y = self.reward_function(
+                reward, self.discount, target, terminal, others
+            )
 
         q = self._criticize(state)
         action_index = self._sampled_action_index(action, q)
@@ -229,6 +231,16 @@ class DQN(TorchFramework):
     def _sampled_action_index(action: Dict[str, t.Tensor], q: t.Tensor):
         idx = action["action"]
         return idx.to(device=q.device, dtype=t.long).view(-1, 1)
+
+    # -- hooks (reference dqn.py:490-501) ------------------------------
+    @staticmethod
+    def action_get_function(sampled_actions: t.Tensor) -> t.Tensor:
+        """Map sampled q-values/indices to the stored action tensor."""
+        return sampled_actions
+
+    @staticmethod
+    def reward_function(reward, discount, next_value, terminal, _):
+        return reward + discount * (1.0 - terminal) * next_value
 
     def update_lr_scheduler(self):
         if self.qnet_lr_sch is not None:

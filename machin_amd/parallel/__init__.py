@@ -1,4 +1,5 @@
 from .event import AndEvent, Event, OrEvent
+from .exception import ExceptionWithTraceback, RemoteTraceback
 from .pickle import Pickler, dumps, loads, mark_static_module
 from .pool import CtxPool, CtxThreadPool, P2PPool, Pool, ThreadPool
 from .process import Process, ProcessException
@@ -27,6 +28,8 @@ __all__ = [
     "OrEvent",
     "AndEvent",
     "Finalize",
+    "ExceptionWithTraceback",
+    "RemoteTraceback",
 ]
 
 

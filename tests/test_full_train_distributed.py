@@ -115,7 +115,7 @@ class TestA3CFullTrain:
             group.barrier()
             return solved
 
-        results = run_multi(fn, timeout=600)
+        results = run_multi(fn, timeout=900)
         assert any(results), "A3C did not solve CartPole on any process"
 
 

@@ -183,3 +183,22 @@ class TestPools:
         with CtxThreadPool(processes=2) as pool:
             out = pool.map(lambda ctx, x: x + ctx, [10, 10])
             assert all(v in (10, 11) for v in out)
+
+
+class TestExceptionTransport:
+    def test_pickled_traceback(self):
+        import pickle
+
+        from machin_amd.parallel.exception import (
+            ExceptionWithTraceback,
+            RemoteTraceback,
+        )
+
+        try:
+            raise ValueError("worker boom")
+        except ValueError as e:
+            data = pickle.dumps(ExceptionWithTraceback(e))
+        exc = pickle.loads(data)
+        assert isinstance(exc, ValueError)
+        assert isinstance(exc.__cause__, RemoteTraceback)
+        assert "worker boom" in str(exc.__cause__)

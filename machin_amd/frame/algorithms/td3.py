@@ -166,7 +166,11 @@ class TD3(DDPG):
         )
 
     def load(self, model_dir, network_map=None, version=-1):
-        TD3.__mro__[2].load(self, model_dir, network_map, version)  # base
+        # skip DDPG.load (it only re-syncs actor/critic): restore all
+        # three target nets, then hard-update the online nets
+        from .base import TorchFramework
+
+        TorchFramework.load(self, model_dir, network_map, version)
         with t.no_grad():
             hard_update(self.actor, self.actor_target)
             hard_update(self.critic, self.critic_target)

@@ -232,6 +232,13 @@ def main():
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
         os.environ.setdefault("MASTER_PORT", "29511")
         dist.init_process_group("nccl", rank=rank, world_size=world_size)
+    if not distributed and args.gpus > 1:
+        raise SystemExit(
+            "--gpus N>1 requires torchrun (one rank per GPU): "
+            "python -m torch.distributed.run --nnodes=1 "
+            f"--nproc-per-node {args.gpus} --master-addr 127.0.0.1 "
+            "bench.py ..."
+        )
     n_gpus = world_size if distributed else args.gpus
 
     t.cuda.set_device(local_rank)

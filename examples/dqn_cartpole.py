@@ -1,0 +1,70 @@
+"""DQN on CartPole (reference analog:
+examples/framework_examples/dqn.py)."""
+import os
+import sys
+
+sys.path.insert(
+    0, os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+)
+
+import argparse
+
+import torch as t
+import torch.nn as nn
+
+from machin_amd.env.envs import CartPoleEnv
+from machin_amd.frame.algorithms import DQN
+
+
+class QNet(nn.Module):
+    def __init__(self, state_dim=4, action_num=2):
+        super().__init__()
+        self.fc1 = nn.Linear(state_dim, 64)
+        self.fc2 = nn.Linear(64, 64)
+        self.fc3 = nn.Linear(64, action_num)
+
+    def forward(self, state):
+        a = t.relu(self.fc1(state))
+        return self.fc3(t.relu(self.fc2(a)))
+
+
+def main(device="cpu", max_episodes=600):
+    dqn = DQN(
+        QNet().to(device), QNet().to(device), t.optim.Adam, nn.MSELoss(),
+        batch_size=64, learning_rate=1e-3, epsilon_decay=0.995,
+        update_rate=0.01, replay_device=device,
+    )
+    env = CartPoleEnv(seed=0)
+    smoothed = 0.0
+    for episode in range(max_episodes):
+        obs = t.tensor(env.reset(), device=device).view(1, 4)
+        total, transitions, done = 0.0, [], False
+        while not done:
+            with t.no_grad():
+                action = dqn.act_discrete_with_noise({"state": obs})
+            o, r, done, _ = env.step(int(action.item()))
+            o = t.tensor(o, device=device).view(1, 4)
+            total += r
+            transitions.append(
+                {"state": {"state": obs}, "action": {"action": action},
+                 "next_state": {"state": o}, "reward": r,
+                 "terminal": done and env.steps < env.max_episode_steps}
+            )
+            obs = o
+        dqn.store_episode(transitions)
+        if dqn.replay_buffer.size() > 500:
+            for _ in range(min(len(transitions), 50)):
+                dqn.update()
+        smoothed = smoothed * 0.9 + total * 0.1
+        if episode % 20 == 0:
+            print(f"episode {episode}: smoothed reward {smoothed:.1f}")
+        if smoothed > 195:
+            print(f"solved at episode {episode}")
+            return
+    print("did not reach 195 within the budget")
+
+
+if __name__ == "__main__":
+    p = argparse.ArgumentParser()
+    p.add_argument("--device", default="cpu")
+    main(p.parse_args().device)

@@ -126,8 +126,6 @@ class GAIL(TorchFramework):
             d["reward"] = float(-t.log(1.0 - prob.clamp(0.0, 0.999)).item())
             shaped.append(d)
         self.cpo.store_episode(shaped)
-        # keep generated pairs for the discriminator update
-        self._last_generated = shaped
 
     # ------------------------------------------------------------------
     def update(self, update_discriminator=True, update_policy=True, **__):

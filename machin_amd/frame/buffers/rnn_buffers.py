@@ -182,12 +182,6 @@ class RNNDistributedBuffer(_RNNWindowMixin, DistributedBuffer):
                                    **kwargs)
         self._init_rnn(sample_length, sample_dimension)
 
-    def sample_batch(self, batch_size, *args, **kwargs):
-        bs, result = DistributedBuffer.sample_batch(
-            self, batch_size, *args, **kwargs
-        )
-        return bs, result
-
     def post_process_batch(self, batch, device, concatenate, sample_attrs,
                            additional_concat_custom_attrs):
         # batch arrives as a flat union of windows from members

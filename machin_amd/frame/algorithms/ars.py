@@ -226,7 +226,7 @@ class ARS(TorchFramework):
         self._generate_deltas()
 
         # all members start from the first member's parameters
-        self._sync_actor(initial=True)
+        self._sync_actor()
 
     # ------------------------------------------------------------------
     @property
@@ -390,7 +390,7 @@ class ARS(TorchFramework):
         self._generate_deltas()
         return True
 
-    def _sync_actor(self, initial: bool = False):
+    def _sync_actor(self):
         members = self.ars_group.get_group_members()
         me = self.ars_group.get_cur_name()
         if me == members[0]:

@@ -60,7 +60,10 @@ class TestConfigGeneration:
             ("RAINBOW", "CartPole-v1"),
             ("A2C", "CartPole-v1"),
             ("PPO", "CartPole-v1"),
+            ("TRPO", "CartPole-v1"),
             ("DDPG", "Pendulum-v1"),
+            ("HDDPG", "Pendulum-v1"),
+            ("DDPGPer", "Pendulum-v1"),
             ("TD3", "Pendulum-v1"),
             ("SAC", "Pendulum-v1"),
         ],

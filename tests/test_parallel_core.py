@@ -202,3 +202,26 @@ class TestExceptionTransport:
         assert isinstance(exc, ValueError)
         assert isinstance(exc.__cause__, RemoteTraceback)
         assert "worker boom" in str(exc.__cause__)
+
+
+class TestMultiP2PQueue:
+    def test_round_robin(self):
+        from machin_amd.parallel.queue import MultiP2PQueue
+
+        q = MultiP2PQueue(3)
+        for i in range(6):
+            q.put(i)
+        got = sorted(q.get(timeout=1) for _ in range(6))
+        assert got == [0, 1, 2, 3, 4, 5]
+        # sub-queue access
+        q.put("x")
+        assert q.get_sub_queue(0).get(timeout=1) == "x"
+        q.close()
+
+    def test_get_timeout(self):
+        from machin_amd.parallel.queue import Empty, MultiP2PQueue
+
+        q = MultiP2PQueue(2)
+        with pytest.raises(Empty):
+            q.get(timeout=0.05)
+        q.close()

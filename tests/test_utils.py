@@ -221,3 +221,79 @@ class TestTracing:
             return a * 2
 
         assert f(21) == 42
+
+
+class TestMediaLogger:
+    def test_local_media_logger(self, tmp_path):
+        from machin_amd.auto.pl_logger import LocalMediaLogger
+
+        img_dir = tmp_path / "img"
+        art_dir = tmp_path / "art"
+        logger = LocalMediaLogger(str(img_dir), str(art_dir))
+        logger.log_metrics({"reward": 1.5}, step=3)
+        src = tmp_path / "thing.txt"
+        src.write_text("data")
+        logger.log_artifact(str(src))
+        logger.finalize()
+        assert (art_dir / "thing.txt").exists()
+        assert "reward" in (art_dir / "metrics.jsonl").read_text()
+
+
+class TestDataset:
+    def test_determine_precision(self):
+        import torch.nn as nn
+
+        from machin_amd.auto.dataset import determine_precision
+
+        assert determine_precision([nn.Linear(2, 2)]) == t.float32
+        mixed = nn.Linear(2, 2)
+        half = nn.Linear(2, 2).half()
+        with pytest.raises(RuntimeError):
+            determine_precision([mixed, half])
+
+    def test_dataset_result(self):
+        from machin_amd.auto.dataset import DatasetResult
+
+        r = DatasetResult()
+        r.add_observation([1, 2])
+        r.add_log({"total_reward": 5})
+        assert len(r) == 1 and r.logs[0]["total_reward"] == 5
+
+
+class TestSpace:
+    def test_discrete_and_box(self):
+        from machin_amd.env.envs.classic_control import Space
+
+        d = Space(n=4, seed=0)
+        assert d.discrete
+        assert all(0 <= d.sample() < 4 for _ in range(20))
+        b = Space(shape=(2,), low=np.array([-1.0, 0.0]),
+                  high=np.array([1.0, 2.0]), seed=0)
+        s = b.sample()
+        assert s.shape == (2,)
+        assert -1 <= s[0] <= 1 and 0 <= s[1] <= 2
+
+
+class TestSaveEnvCleanup:
+    def test_remove_old_trials(self, tmp_path):
+        import time as _time
+
+        env = SaveEnv(str(tmp_path))
+        # fabricate an old trial directory (2 hours ago)
+        old_name = _time.strftime(
+            "%Y_%m_%d_%H_%M_%S", _time.localtime(_time.time() - 7200)
+        )
+        os.makedirs(tmp_path / old_name)
+        env.remove_trials_older_than(diff_hour=1)
+        assert not (tmp_path / old_name).exists()
+        assert os.path.isdir(env.get_trial_root())
+
+
+class TestMediaSubproc:
+    def test_image_subproc(self, tmp_path):
+        from machin_amd.utils.media import create_image_subproc
+
+        img = np.random.rand(8, 8, 3).astype(np.float32)
+        wait = create_image_subproc(img, str(tmp_path), "async_frame")
+        wait()
+        assert os.path.exists(tmp_path / "async_frame.png")

@@ -131,7 +131,8 @@ class DQNApex(DQNPer):
             if world.rank in learner_ranks and learner_n > 1 else None
         if world.rank in learner_ranks and learner_n > 1:
             models[0] = DistributedDataParallel(
-                models[0], process_group=coll.group
+                models[0], process_group=coll.group,
+                reduction=fc.get("ddp_reduction", "all_reduce"),
             )
         optimizer = assert_and_get_valid_optimizer(fc["optimizer"])
         criterion = assert_and_get_valid_criterion(fc["criterion"])(
@@ -148,13 +149,20 @@ class DQNApex(DQNPer):
                     "optimizer", "criterion", "criterion_args",
                     "criterion_kwargs", "lr_scheduler",
                     "apex_group_name", "apex_group_members",
-                    "learner_process_number",
+                    "learner_process_number", "ddp_reduction",
                 )
             },
         )
         if world.rank not in learner_ranks:
             # sampler ranks never update
             frame.update = lambda *a, **k: 0.0
+        else:
+            # wire reducer finalize + bucket-preserving zero_grad into
+            # the framework's pluggable backward (no-op when no model
+            # is DDP-wrapped)
+            from ...parallel.ddp import install_ddp_finalize
+
+            install_ddp_finalize(frame)
         return frame
 
 
@@ -248,3 +256,88 @@ class DDPGApex(DDPGPer):
                 getattr(self.actor, "module", self.actor)
             )
         return result
+
+    @classmethod
+    def generate_config(cls, config):
+        config = DDPGPer.generate_config(config)
+        config["frame"] = "DDPGApex"
+        fc = config["frame_config"]
+        fc["frame"] = "DDPGApex"
+        fc.setdefault("apex_group_name", "apex_group")
+        fc.setdefault("apex_group_members", "all")
+        fc.setdefault("learner_process_number", 1)
+        return config
+
+    @classmethod
+    def init_from_config(cls, config, model_device="cpu"):
+        from ...frame.helpers.servers import model_server_helper
+        from ...parallel.ddp import (
+            DistributedDataParallel,
+            install_ddp_finalize,
+        )
+        from ...parallel.distributed.world import get_world
+        from .utils import (
+            assert_and_get_valid_criterion,
+            assert_and_get_valid_models,
+            assert_and_get_valid_optimizer,
+        )
+
+        data = config.data if hasattr(config, "data") else dict(config)
+        fc = data["frame_config"]
+        world = get_world()
+        servers = model_server_helper(model_num=1)
+        members = (
+            world.get_members()
+            if fc.get("apex_group_members", "all") == "all"
+            else fc["apex_group_members"]
+        )
+        group = world.create_rpc_group(
+            fc.get("apex_group_name", "apex_group"), members
+        )
+        model_cls = assert_and_get_valid_models(fc["models"])
+        models = [
+            m(*args, **kwargs).to(model_device)
+            for m, args, kwargs in zip(
+                model_cls, fc.get("model_args", ((), (), (), ())),
+                fc.get("model_kwargs", ({}, {}, {}, {})),
+            )
+        ]
+        learner_n = fc.get("learner_process_number", 1)
+        learner_ranks = list(range(learner_n))
+        coll = world.create_collective_group(learner_ranks) \
+            if world.rank in learner_ranks and learner_n > 1 else None
+        if coll is not None:
+            # learner ranks sync actor (models[0]) and critic
+            # (models[2]) gradients; targets follow via soft_update
+            reduction = fc.get("ddp_reduction", "all_reduce")
+            models[0] = DistributedDataParallel(
+                models[0], process_group=coll.group, reduction=reduction
+            )
+            models[2] = DistributedDataParallel(
+                models[2], process_group=coll.group, reduction=reduction
+            )
+        optimizer = assert_and_get_valid_optimizer(fc["optimizer"])
+        criterion = assert_and_get_valid_criterion(fc["criterion"])(
+            *fc.get("criterion_args", ()), **fc.get("criterion_kwargs", {})
+        )
+        frame = cls(
+            models[0], models[1], models[2], models[3],
+            optimizer, criterion, group, servers,
+            **{
+                k: v
+                for k, v in fc.items()
+                if k
+                not in (
+                    "frame", "models", "model_args", "model_kwargs",
+                    "optimizer", "criterion", "criterion_args",
+                    "criterion_kwargs", "lr_scheduler",
+                    "apex_group_name", "apex_group_members",
+                    "learner_process_number", "ddp_reduction",
+                )
+            },
+        )
+        if world.rank not in learner_ranks:
+            frame.update = lambda *a, **k: (0.0, 0.0)
+        else:
+            install_ddp_finalize(frame)
+        return frame

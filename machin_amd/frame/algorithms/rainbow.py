@@ -102,18 +102,20 @@ class RAINBOW(DQNPer):
                  else float(tr.terminal.reshape(-1)[0]) for tr in episode]
             )
             nstep = ops.nstep_returns(rewards, terminals, self.discount, n)
+            # single host copy for the whole episode (no per-element
+            # .item() synchronization)
+            nstep_list = nstep.tolist()
+            term_list = terminals.tolist()
             new_episode = []
             for i, tr in enumerate(episode):
                 d = {k: getattr(tr, k) for k in tr.keys()}
-                d["reward"] = float(nstep[i].item())
+                d["reward"] = float(nstep_list[i])
                 # bootstrap state is s_{i+n} (or the episode end)
                 j = min(i + n - 1, T - 1)
                 d["next_state"] = {
                     k: v for k, v in episode[j].next_state.items()
                 }
-                d["terminal"] = bool(
-                    terminals[i : j + 1].max().item() > 0.5
-                )
+                d["terminal"] = bool(max(term_list[i : j + 1]) > 0.5)
                 new_episode.append(Transition(**d))
             episode = new_episode
         self.replay_buffer.store_episode(

@@ -36,6 +36,10 @@ def soft_update(target_net: nn.Module, source_net: nn.Module, update_rate: float
 
 def hard_update(target_net: nn.Module, source_net: nn.Module):
     """Copy every parameter of ``source_net`` into ``target_net``."""
+    # unwrap DDP containers: a wrapped source's state_dict carries
+    # "module."-prefixed keys that would not match a bare target
+    target_net = _unwrap(target_net)
+    source_net = _unwrap(source_net)
     target_net.load_state_dict(source_net.state_dict())
 
 

@@ -14,8 +14,18 @@ explicit reducer designed for one node of 8×MI355X:
 * bucket size defaults to 32 MiB: xGMI is 7 point-to-point links at
   ~153 GB/s, ring all-reduce is per-link bound, so fewer, larger
   collectives beat DDP's default small buckets for these model sizes;
-* no autograd graph rewriting, no "find_unused_parameters" machinery —
-  call :meth:`finalize` after backward, and use :meth:`zero_grad_`.
+* ``reduction="reduce_scatter"`` splits each bucket into a
+  reduce-scatter + all-gather pair (half the per-link bytes of a ring
+  all-reduce on xGMI for large buckets); falls back to all-reduce on
+  backends without reduce_scatter support (gloo);
+* no autograd graph rewriting, no "find_unused_parameters" machinery.
+
+Integration contract: after ``backward()`` call :meth:`finalize` (or
+let :func:`install_ddp_finalize` hook it into a framework's pluggable
+backward), and zero gradients with :meth:`zero_grad_`. Calling
+``optimizer.zero_grad(set_to_none=True)`` anyway is SAFE but slower:
+the post-accumulate hook detects the detached grad, copies it back
+into the bucket and rebinds the view.
 """
 from typing import List, Optional
 
@@ -33,12 +43,21 @@ class GradReducer:
         process_group: Optional[dist.ProcessGroup] = None,
         bucket_cap_mb: float = 32.0,
         average: bool = True,
+        reduction: str = "all_reduce",
     ):
+        if reduction not in ("all_reduce", "reduce_scatter"):
+            raise ValueError(
+                "reduction must be 'all_reduce' or 'reduce_scatter'."
+            )
         self.module = module
         self.group = process_group
         self.average = average
+        self.reduction = reduction
         self._works: List = []
         self._hooks = []
+        self.world_size = (
+            dist.get_world_size(self.group) if dist.is_initialized() else 1
+        )
 
         params = [p for p in module.parameters() if p.requires_grad]
         if not params:
@@ -61,14 +80,26 @@ class GradReducer:
 
         for bucket in self.buckets:
             total = sum(p.numel() for p in bucket["params"])
-            flat = t.zeros(total, dtype=params[0].dtype, device=device)
+            padded = total
+            if self.reduction == "reduce_scatter" and self.world_size > 1:
+                ws = self.world_size
+                padded = (total + ws - 1) // ws * ws
+            flat = t.zeros(padded, dtype=params[0].dtype, device=device)
             bucket["flat"] = flat
             bucket["ready"] = 0
+            bucket["views"] = []
+            bucket["view_of"] = {}
+            if padded != total or self.reduction == "reduce_scatter":
+                bucket["rs_out"] = t.zeros(
+                    padded // max(self.world_size, 1),
+                    dtype=params[0].dtype, device=device,
+                )
             offset = 0
             for p in bucket["params"]:
                 view = flat[offset : offset + p.numel()].view_as(p)
                 p.grad = view
-                bucket["views"] = bucket.get("views", []) + [view]
+                bucket["views"].append(view)
+                bucket["view_of"][id(p)] = view
                 offset += p.numel()
 
         self._param_bucket = {}
@@ -80,37 +111,77 @@ class GradReducer:
             h = p.register_post_accumulate_grad_hook(self._on_grad_ready)
             self._hooks.append(h)
 
-        self.world_size = (
-            dist.get_world_size(self.group) if dist.is_initialized() else 1
-        )
-
     # ------------------------------------------------------------------
+    def _backend_has_reduce_scatter(self) -> bool:
+        try:
+            backend = dist.get_backend(self.group)
+        except Exception:  # noqa: BLE001 - conservative fallback
+            return False
+        return str(backend) in ("nccl", "rccl")
+
     def _on_grad_ready(self, param: t.Tensor):
         bi = self._param_bucket[id(param)]
         bucket = self.buckets[bi]
+        # survive optimizer.zero_grad(set_to_none=True): autograd then
+        # allocated a FRESH grad tensor — fold it into the bucket and
+        # rebind the view so the reduction sees the real gradient
+        view = bucket["view_of"][id(param)]
+        if param.grad is not view:
+            with t.no_grad():
+                view.copy_(param.grad)
+            param.grad = view
         bucket["ready"] += 1
         if bucket["ready"] == len(bucket["params"]):
             bucket["ready"] = 0
             if self.world_size > 1:
-                work = dist.all_reduce(
-                    bucket["flat"], op=dist.ReduceOp.SUM,
-                    group=self.group, async_op=True,
-                )
-                self._works.append(work)
+                if (
+                    self.reduction == "reduce_scatter"
+                    and self._backend_has_reduce_scatter()
+                ):
+                    work = dist.reduce_scatter_tensor(
+                        bucket["rs_out"], bucket["flat"],
+                        op=dist.ReduceOp.SUM,
+                        group=self.group, async_op=True,
+                    )
+                    self._works.append(("rs", bi, work))
+                else:
+                    work = dist.all_reduce(
+                        bucket["flat"], op=dist.ReduceOp.SUM,
+                        group=self.group, async_op=True,
+                    )
+                    self._works.append(("ar", bi, work))
 
     def finalize(self):
-        """Wait for all in-flight reductions; call after backward()."""
-        for work in self._works:
+        """Wait for all in-flight reductions; call after backward().
+
+        Only buckets that were actually reduced in this backward pass
+        are averaged — a partial backward (e.g. a loss touching one of
+        two wrapped models) must not rescale stale buckets.
+        """
+        reduced_flats = []
+        gather_buckets = []
+        for kind, bi, work in self._works:
             work.wait()
+            reduced_flats.append(self.buckets[bi]["flat"])
+            if kind == "rs":
+                gather_buckets.append(bi)
         self._works.clear()
-        if self.world_size > 1 and self.average:
-            t._foreach_mul_(
-                [b["flat"] for b in self.buckets], 1.0 / self.world_size
-            )
+        if gather_buckets:
+            gathers = [
+                dist.all_gather_into_tensor(
+                    self.buckets[bi]["flat"], self.buckets[bi]["rs_out"],
+                    group=self.group, async_op=True,
+                )
+                for bi in gather_buckets
+            ]
+            for w in gathers:
+                w.wait()
+        if self.world_size > 1 and self.average and reduced_flats:
+            t._foreach_mul_(reduced_flats, 1.0 / self.world_size)
 
     def zero_grad_(self):
-        """Zero the flat buffers (keeps the grad views intact — do NOT
-        use optimizer.zero_grad(set_to_none=True))."""
+        """Zero the flat buffers (keeps the grad views intact — the
+        fast equivalent of optimizer.zero_grad())."""
         t._foreach_zero_([b["flat"] for b in self.buckets])
 
     def rebind_grads(self):
@@ -130,20 +201,29 @@ class DistributedDataParallel(nn.Module):
 
     API-compatible with the reference's learner wrapping: ``forward``
     delegates, ``.module`` exposes the inner net. After ``backward()``
-    call ``.finalize()`` (or use :meth:`sync_context`)."""
+    call ``.finalize()`` (or install :func:`install_ddp_finalize` on
+    the owning framework)."""
 
     def __init__(self, module: nn.Module, process_group=None,
-                 bucket_cap_mb: float = 32.0):
+                 bucket_cap_mb: float = 32.0,
+                 reduction: str = "all_reduce"):
         super().__init__()
         self.module = module
-        self.reducer = GradReducer(module, process_group, bucket_cap_mb)
+        self.reducer = GradReducer(
+            module, process_group, bucket_cap_mb, reduction=reduction
+        )
         # broadcast initial params so all ranks start identical
         if self.reducer.world_size > 1:
+            src = (
+                dist.get_global_rank(process_group, 0)
+                if process_group is not None
+                else 0
+            )
             with t.no_grad():
                 for p in module.parameters():
-                    dist.broadcast(p.data, src=0, group=process_group)
+                    dist.broadcast(p.data, src=src, group=process_group)
                 for b in module.buffers():
-                    dist.broadcast(b.data, src=0, group=process_group)
+                    dist.broadcast(b.data, src=src, group=process_group)
 
     def forward(self, *args, **kwargs):
         return self.module(*args, **kwargs)
@@ -153,3 +233,65 @@ class DistributedDataParallel(nn.Module):
 
     def zero_grad_(self):
         self.reducer.zero_grad_()
+
+
+def install_ddp_finalize(frame):
+    """Wire a framework's pluggable backward to its GradReducers.
+
+    Fixes the multi-learner integration contract in one place instead
+    of per-algorithm: after this call,
+
+    * every ``frame._backward(loss)`` runs autograd backward and then
+      ``finalize()``s each wrapped model's reducer (awaits + averages
+      the async all-reduces) BEFORE the algorithm calls
+      ``optimizer.step()``;
+    * each optimizer whose parameters are fully covered by one wrapped
+      model zeroes the flat bucket buffers instead of detaching the
+      grad views on ``zero_grad(set_to_none=True)``.
+
+    Reference analog: torch DDP does this inside its own backward
+    hooks (reference machin/frame/algorithms/apex.py:213-221 relies on
+    torch DDP); our reducer keeps it explicit.
+    """
+    wrapped = []
+    for v in frame.__dict__.values():
+        if isinstance(v, DistributedDataParallel) and v not in wrapped:
+            wrapped.append(v)
+    if not wrapped:
+        return frame
+    reducers = [m.reducer for m in wrapped]
+
+    orig_backward = frame.backward_function
+
+    def ddp_backward(*args, **kwargs):
+        orig_backward(*args, **kwargs)
+        for r in reducers:
+            r.finalize()
+
+    frame.set_backward_function(ddp_backward)
+
+    # optimizer.zero_grad must not detach bucket views
+    covered = {}
+    for r in reducers:
+        for b in r.buckets:
+            for p in b["params"]:
+                covered[id(p)] = r
+    try:
+        optimizers = frame.optimizers
+    except (NotImplementedError, AttributeError):
+        optimizers = []
+    for opt in optimizers or []:
+        opt_params = [
+            p for g in opt.param_groups for p in g["params"]
+            if p.requires_grad
+        ]
+        opt_reducers = {
+            covered[id(p)] for p in opt_params if id(p) in covered
+        }
+        if opt_reducers and all(id(p) in covered for p in opt_params):
+            def _zero_grad(set_to_none=True, _rs=tuple(opt_reducers)):
+                for r in _rs:
+                    r.zero_grad_()
+
+            opt.zero_grad = _zero_grad
+    return frame

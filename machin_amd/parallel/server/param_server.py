@@ -277,16 +277,18 @@ class PushPullGradServerImpl:
         return out
 
     def _apply(self, grads: Dict[str, t.Tensor], count: int):
+        # grads arriving here are already reduced by _reduce_batch
+        # (which divides by the batch count when reduce_method is
+        # "mean") — scaling again here would shrink applied gradients
+        # by ~count×. ``count`` is kept for the
+        # service signature only.
         with self._model_lock:
             if self.model is None:
                 return
             params = dict(self.model.named_parameters())
             for k, g in grads.items():
                 if k in params:
-                    scale = (
-                        1.0 / count if self.reduce_method == "mean" else 1.0
-                    )
-                    params[k].grad = g.to(params[k].device) * scale
+                    params[k].grad = g.to(params[k].device)
             self.optimizer.step()
             self.optimizer.zero_grad(set_to_none=False)
             self._publish()

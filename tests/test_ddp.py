@@ -102,3 +102,97 @@ class TestGradReducer:
             return True
 
         assert all(run_multi(fn, world_size=2))
+
+
+class TestGradReducerRobustness:
+    def test_survives_set_to_none_zero_grad(self):
+        """optimizer.zero_grad(set_to_none=True) detaches the bucket
+        views; the post-accumulate hook must fold the fresh grads back
+        in so reduction still happens (round-1 ADVICE high)."""
+        def fn(rank, world):
+            import torch.nn as nn
+
+            from machin_amd.parallel.ddp import GradReducer
+
+            t.manual_seed(0)
+            model = nn.Linear(8, 4)
+            reducer = GradReducer(model)
+            opt = t.optim.SGD(model.parameters(), lr=0.1)
+            t.manual_seed(100 + rank)
+            x = t.rand(16, 8)
+            opt.zero_grad(set_to_none=True)  # detaches bucket views
+            model(x).sum().backward()
+            reducer.finalize()
+            # grads must be identical (averaged) across ranks
+            import torch.distributed as dist
+
+            g = model.weight.grad.flatten().clone()
+            out = [t.zeros_like(g) for _ in range(world.world_size)]
+            dist.all_gather(out, g)
+            assert t.allclose(out[0], out[1], atol=1e-6)
+            # and the views must be re-bound for the optimizer step
+            assert model.weight.grad is reducer.buckets[
+                reducer._param_bucket[id(model.weight)]
+            ]["view_of"][id(model.weight)]
+            return True
+
+        assert all(run_multi(fn, world_size=2))
+
+    def test_reduce_scatter_mode(self):
+        """reduce_scatter mode: averaged numerics on any backend
+        (falls back to all-reduce on gloo, real RS+AG on RCCL)."""
+        def fn(rank, world):
+            import torch.nn as nn
+
+            from machin_amd.parallel.ddp import GradReducer
+
+            t.manual_seed(0)
+            model = nn.Sequential(nn.Linear(8, 16), nn.Linear(16, 3))
+            reducer = GradReducer(model, reduction="reduce_scatter")
+            t.manual_seed(200 + rank)
+            x = t.rand(4, 8)
+            reducer.zero_grad_()
+            model(x).sum().backward()
+            reducer.finalize()
+            import torch.distributed as dist
+
+            g = t.cat([p.grad.flatten() for p in model.parameters()])
+            out = [t.zeros_like(g) for _ in range(world.world_size)]
+            dist.all_gather(out, g)
+            assert t.allclose(out[0], out[1], atol=1e-6)
+            return True
+
+        assert all(run_multi(fn, world_size=2))
+
+    def test_matches_torch_ddp(self):
+        """GradReducer's reduced grads equal torch-DDP's on the same
+        model + per-rank data (round-1 VERDICT next #2)."""
+        def fn(rank, world):
+            import torch.nn as nn
+
+            from machin_amd.parallel.ddp import GradReducer
+
+            t.manual_seed(7)
+            model = nn.Sequential(
+                nn.Linear(8, 32), nn.ReLU(), nn.Linear(32, 4)
+            )
+            t.manual_seed(7)
+            ref = nn.Sequential(
+                nn.Linear(8, 32), nn.ReLU(), nn.Linear(32, 4)
+            )
+            ref = nn.parallel.DistributedDataParallel(ref)
+            reducer = GradReducer(model)
+            t.manual_seed(300 + rank)
+            x = t.rand(16, 8)
+            y = t.rand(16, 4)
+            reducer.zero_grad_()
+            ((model(x) - y) ** 2).mean().backward()
+            reducer.finalize()
+            ((ref(x) - y) ** 2).mean().backward()
+            for p, q in zip(model.parameters(), ref.parameters()):
+                assert t.allclose(p.grad, q.grad, atol=1e-6), (
+                    f"grad mismatch vs torch DDP on rank {rank}"
+                )
+            return True
+
+        assert all(run_multi(fn, world_size=2, timeout=240))

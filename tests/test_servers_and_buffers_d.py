@@ -131,6 +131,37 @@ class TestGradServer:
         results = run_multi(fn)
         assert all(results)
 
+    def test_mean_reduce_magnitude(self):
+        """reduce_method="mean" must divide by the batch count exactly
+        once (round-1 ADVICE medium: _apply scaled a second time).
+        With SGD lr=1 and a batch of three ones-gradients, the applied
+        delta must be exactly -1*ones, not -1/3."""
+        import threading
+
+        from machin_amd.parallel.server.param_server import (
+            PushPullGradServerImpl,
+        )
+
+        for method, expect in (("mean", -1.0), ("sum", -3.0)):
+            net = _small_net()
+            srv = PushPullGradServerImpl.__new__(PushPullGradServerImpl)
+            srv.reduce_method = method
+            srv.model = net
+            srv.optimizer = t.optim.SGD(net.parameters(), lr=1.0)
+            srv._model_lock = threading.Lock()
+            srv._publish = lambda: None
+            before = net.weight.detach().clone()
+            batch = [
+                {k: t.ones_like(v) for k, v in net.named_parameters()}
+                for _ in range(3)
+            ]
+            reduced = srv._reduce_batch(batch)
+            srv._apply(reduced, len(batch))
+            delta = net.weight.detach() - before
+            assert t.allclose(
+                delta, expect * t.ones_like(delta), atol=1e-6
+            ), f"{method}: delta {delta.flatten()[0]} != {expect}"
+
 
 class TestDistributedBuffer:
     def test_global_sampling(self):

@@ -25,6 +25,36 @@ from .ddpg_per import DDPGPer
 from .dqn_per import DQNPer
 
 
+def _resolve_device_replay(fc, world, learner_n, model_device):
+    """When the config asks for cuda replay, resolve the per-rank
+    shard device (each learner rank owns its own GPU) and name the
+    learner members that host HBM shards."""
+    rd = fc.get("replay_device", "cpu")
+    if t.device(rd).type == "cuda":
+        if t.device(model_device).type == "cuda":
+            fc["replay_device"] = model_device
+        fc["replay_learners"] = world.get_members()[:learner_n]
+    return fc
+
+
+def _apex_buffer(name, group, replay_size, replay_device,
+                 replay_learners):
+    """Pick the replay backend: learner-resident HBM shards
+    (store-side push, sampling local to the learner GPU) when the
+    replay device is cuda or learner shards are named; otherwise the
+    reference-style sample-side-pull CPU buffer."""
+    if t.device(replay_device).type == "cuda" or replay_learners:
+        from ..buffers.device_buffer_d import (
+            DeviceDistributedPrioritizedBuffer,
+        )
+
+        return DeviceDistributedPrioritizedBuffer(
+            name, group, replay_size, learners=replay_learners,
+            device=replay_device,
+        )
+    return DistributedPrioritizedBuffer(name, group, replay_size)
+
+
 class DQNApex(DQNPer):
     def __init__(
         self,
@@ -36,10 +66,13 @@ class DQNApex(DQNPer):
         model_server: Tuple[PushPullModelServer],
         *_,
         replay_size: int = 500000,
+        replay_device="cpu",
+        replay_learners=None,
         **kwargs,
     ):
-        buffer = DistributedPrioritizedBuffer(
-            "dqn_apex_buffer", apex_group, replay_size
+        buffer = _apex_buffer(
+            "dqn_apex_buffer", apex_group, replay_size, replay_device,
+            replay_learners,
         )
         super().__init__(
             qnet, qnet_target, optimizer, criterion,
@@ -126,6 +159,9 @@ class DQNApex(DQNPer):
         ]
         learner_n = fc.get("learner_process_number", 1)
         learner_ranks = list(range(learner_n))
+        fc = _resolve_device_replay(
+            dict(fc), world, learner_n, model_device
+        )
         # every process must join new_group creation collectively
         coll = world.create_collective_group(learner_ranks) \
             if world.rank in learner_ranks and learner_n > 1 else None
@@ -179,10 +215,13 @@ class DDPGApex(DDPGPer):
         model_server: Tuple[PushPullModelServer],
         *_,
         replay_size: int = 500000,
+        replay_device="cpu",
+        replay_learners=None,
         **kwargs,
     ):
-        buffer = DistributedPrioritizedBuffer(
-            "ddpg_apex_buffer", apex_group, replay_size
+        buffer = _apex_buffer(
+            "ddpg_apex_buffer", apex_group, replay_size, replay_device,
+            replay_learners,
         )
         super().__init__(
             actor, actor_target, critic, critic_target, optimizer,
@@ -304,6 +343,9 @@ class DDPGApex(DDPGPer):
         ]
         learner_n = fc.get("learner_process_number", 1)
         learner_ranks = list(range(learner_n))
+        fc = _resolve_device_replay(
+            dict(fc), world, learner_n, model_device
+        )
         coll = world.create_collective_group(learner_ranks) \
             if world.rank in learner_ranks and learner_n > 1 else None
         if coll is not None:

@@ -10,6 +10,7 @@ import numpy as np
 import torch as t
 import torch.nn as nn
 
+from ..buffers import default_buffer as _default_buffer
 from ..buffers.prioritized_buffer import PrioritizedBuffer
 from .ddpg import DDPG
 from .utils import hard_update, soft_update
@@ -38,7 +39,9 @@ class DDPGPer(DDPG):
             optimizer,
             criterion,
             replay_buffer=(
-                PrioritizedBuffer(replay_size, replay_device)
+                # cuda replay_device -> DeviceSumTree PER in HBM
+                _default_buffer(replay_size, replay_device,
+                                prioritized=True)
                 if replay_buffer is None
                 else replay_buffer
             ),
@@ -93,8 +96,13 @@ class DDPGPer(DDPG):
         ).view(batch_size)
         value_loss = (per_sample * weights).mean()
 
-        abs_td = (cur_value - y).detach().abs().view(batch_size).cpu().numpy()
-        self.replay_buffer.update_priority(abs_td.astype(np.float64), index)
+        abs_td = (cur_value - y).detach().abs().view(batch_size)
+        if getattr(self.replay_buffer, "accepts_tensor_priorities", False):
+            self.replay_buffer.update_priority(abs_td, index)
+        else:
+            self.replay_buffer.update_priority(
+                abs_td.cpu().numpy().astype(np.float64), index
+            )
 
         if self.visualize:
             self.visualize_model(value_loss, "critic", self.visualize_dir)

@@ -90,8 +90,12 @@ class DQN(TorchFramework):
             self.qnet_optim = optimizer
         else:
             self.qnet_optim = optimizer(self.qnet.parameters(), lr=learning_rate)
+        # replay_device="cuda:*" routes replay to the HBM-resident
+        # flat rings (machin_amd/frame/buffers/device_buffer.py)
+        from ..buffers import default_buffer
+
         self.replay_buffer = (
-            Buffer(replay_size, replay_device)
+            default_buffer(replay_size, replay_device)
             if replay_buffer is None
             else replay_buffer
         )

@@ -28,13 +28,17 @@ class DQNPer(DQN):
         **kwargs,
     ):
         kwargs.pop("mode", None)
+        from ..buffers import default_buffer
+
         super().__init__(
             qnet,
             qnet_target,
             optimizer,
             criterion,
             replay_buffer=(
-                PrioritizedBuffer(replay_size, replay_device)
+                # cuda replay_device -> DeviceSumTree PER in HBM
+                default_buffer(replay_size, replay_device,
+                               prioritized=True)
                 if replay_buffer is None
                 else replay_buffer
             ),
@@ -85,13 +89,15 @@ class DQNPer(DQN):
         ).view(batch_size)
         loss = (per_sample * weights).mean()
 
-        # new priorities = |TD error|
-        abs_td = (
-            (q_taken - y).detach().abs().view(batch_size).cpu().numpy()
-        )
-        self.replay_buffer.update_priority(
-            abs_td.astype(np.float64), index
-        )
+        # new priorities = |TD error|; device buffers take the tensor
+        # directly (no D2H sync on the update path)
+        abs_td = (q_taken - y).detach().abs().view(batch_size)
+        if getattr(self.replay_buffer, "accepts_tensor_priorities", False):
+            self.replay_buffer.update_priority(abs_td, index)
+        else:
+            self.replay_buffer.update_priority(
+                abs_td.cpu().numpy().astype(np.float64), index
+            )
 
         if self.visualize:
             self.visualize_model(loss, "qnet", self.visualize_dir)

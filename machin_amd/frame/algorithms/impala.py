@@ -143,6 +143,12 @@ class IMPALA(TorchFramework):
         self.replay_buffer = IMPALABuffer(
             "impala_buffer", impala_group, replay_size
         )
+        # optional shared-memory rollout ring (use_rollout_ring)
+        self._ring = None
+        self._ring_codec = None
+        self._ring_unroll = None
+        self._ring_pinned = None
+        self._ring_timeout = 0.5
         self.actor_model_server = model_server[0]
         self.is_syncing = True
 
@@ -215,6 +221,11 @@ class IMPALA(TorchFramework):
     def store_episode(self, episode: List[Union[Transition, Dict]]):
         """Store one episode; each transition must carry the behavior
         policy's ``action_log_prob`` custom attribute."""
+        if self._ring is not None:
+            # shared-memory path: fixed-shape segments, zero
+            # serialization — only slot indices cross processes
+            self._ring_codec.write_episode(self._ring, episode)
+            return
         transitions = []
         for tr in episode:
             if isinstance(tr, dict):
@@ -236,6 +247,10 @@ class IMPALA(TorchFramework):
         update_policy, update_target) — the actor step is gated on
         ``update_policy`` (reference machin/frame/algorithms/
         impala.py:242-253)."""
+        if self._ring is not None:
+            return self._update_from_ring(
+                update_value, update_policy, update_target
+            )
         episodes = self.replay_buffer.pop_episodes(self.batch_size)
         if not episodes:
             return 0.0, 0.0
@@ -370,6 +385,133 @@ class IMPALA(TorchFramework):
             nn.utils.clip_grad_norm_(self.actor.parameters(), self.grad_max)
             self.actor_optim.step()
 
+        if update_target:
+            self.actor_model_server.push(
+                getattr(self.actor, "module", self.actor)
+            )
+        return (
+            -float(act_policy_loss.detach().item()),
+            float(value_loss.detach().item()),
+        )
+
+    # ------------------------------------------------------------------
+    # shared-memory rollout ring mode (MI355X fast path; no reference
+    # counterpart — replaces pickled-episode RPC, reference
+    # machin/frame/buffers/buffer_d.py:194-197)
+    # ------------------------------------------------------------------
+    def use_rollout_ring(self, ring, drain_timeout: float = 0.5):
+        """Attach a :class:`machin_amd.parallel.rollout_ring.
+        RolloutRing` built by ``make_episode_ring``. Afterwards:
+
+        * actor processes: ``store_episode`` writes fixed-shape
+          segments straight into shared memory (episodes longer than
+          the ring's unroll are split into consecutive segments);
+        * the learner: ``update`` drains ready slots, stages them
+          through a reusable pinned slab, and issues ONE async H2D
+          copy per attribute before the batched V-trace update —
+          actor rollouts land in host DRAM and stream into HBM
+          (BASELINE "async rollout staging").
+        """
+        from ...parallel.rollout_ring import EpisodeSegmentCodec
+
+        self._ring = ring
+        self._ring_unroll = int(ring.spec["reward"][0][0])
+        self._ring_codec = EpisodeSegmentCodec(self._ring_unroll)
+        self._ring_timeout = drain_timeout
+
+    def _update_from_ring(self, update_value, update_policy,
+                          update_target):
+        ring, T = self._ring, self._ring_unroll
+        idx = ring.drain(self.batch_size, timeout=self._ring_timeout)
+        if not idx:
+            return 0.0, 0.0
+        self.actor.train()
+        self.critic.train()
+        device = next(self.critic.parameters()).device
+        if self._ring_pinned is None and device.type == "cuda":
+            self._ring_pinned = ring.make_pinned_staging(
+                max(self.batch_size, 64)
+            )
+        batch = ring.gather(idx, device, pinned=self._ring_pinned)
+        ring.release(idx)
+        B = len(idx)
+
+        lengths = batch["length"].view(B)
+        sT = t.arange(T, device=device).view(T, 1)
+        mask = (sT < lengths.view(1, B)).float()
+        pad_row = (sT == lengths.view(1, B)).float()  # row L (if L<T)
+        rewards = batch["reward"].t() * mask
+        terminals = batch["terminal"].t()
+        blp = batch["behavior_logp"].t() * mask
+        nd = (1.0 - terminals) * mask
+
+        def tmajor(v):
+            # [B, T, ...] -> [T*B, ...] in t-major order
+            return v.transpose(0, 1).reshape(T * B, *v.shape[2:])
+
+        flat_state = {
+            k.split("/", 1)[1]: tmajor(v)
+            for k, v in batch.items() if k.startswith("state/")
+        }
+        flat_action = {
+            k.split("/", 1)[1]: tmajor(v)
+            for k, v in batch.items() if k.startswith("action/")
+        }
+        boot_state = {
+            k.split("/", 1)[1]: v
+            for k, v in batch.items() if k.startswith("boot_state/")
+        }
+
+        result = self._eval_act(flat_state, flat_action)
+        taken_logp = result[1].view(T, B)
+        entropy = result[2].view(T, B) if len(result) > 2 else None
+        values = self._criticize(flat_state).view(T, B)
+
+        with t.no_grad():
+            boot = self._criticize(boot_state).view(B)
+            last_term = terminals.gather(
+                0, (lengths - 1).clamp_min(0).view(1, B)
+            ).view(B)
+            boot = boot * (1.0 - last_term)
+            det_values = values.detach() * mask \
+                + boot.view(1, B) * pad_row
+            rew_pad = rewards + boot.view(1, B) * pad_row
+            vs, pg_adv = ops.vtrace(
+                blp,
+                taken_logp.detach() * mask,
+                rew_pad,
+                det_values,
+                boot,
+                1.0 - nd,
+                self.discount,
+                rho_clip=self.isw_clip_rho,
+                c_clip=self.isw_clip_c,
+                pg_rho_clip=self.isw_clip_rho,
+            )
+
+        n_valid = mask.sum().clamp_min(1.0)
+        act_policy_loss = -(pg_adv * taken_logp * mask).sum() / n_valid
+        if self.entropy_weight is not None and entropy is not None:
+            act_policy_loss = act_policy_loss - self.entropy_weight * (
+                (entropy * mask).sum() / n_valid
+            )
+        value_elems = self.criterion(values * mask, vs * mask)
+        value_loss = (value_elems.sum() / n_valid) * self.value_weight
+
+        if update_value:
+            self.critic_optim.zero_grad(set_to_none=True)
+            self._backward(value_loss, retain_graph=update_policy)
+            nn.utils.clip_grad_norm_(
+                self.critic.parameters(), self.grad_max
+            )
+            self.critic_optim.step()
+        if update_policy:
+            self.actor_optim.zero_grad(set_to_none=True)
+            self._backward(act_policy_loss)
+            nn.utils.clip_grad_norm_(
+                self.actor.parameters(), self.grad_max
+            )
+            self.actor_optim.step()
         if update_target:
             self.actor_model_server.push(
                 getattr(self.actor, "module", self.actor)

@@ -167,9 +167,12 @@ class RAINBOW(DQNPer):
         ).view(B)
         loss = (per_sample * weights).mean()
 
-        self.replay_buffer.update_priority(
-            per_sample.detach().cpu().numpy().astype(np.float64), index
-        )
+        if getattr(self.replay_buffer, "accepts_tensor_priorities", False):
+            self.replay_buffer.update_priority(per_sample.detach(), index)
+        else:
+            self.replay_buffer.update_priority(
+                per_sample.detach().cpu().numpy().astype(np.float64), index
+            )
 
         if self.visualize:
             self.visualize_model(loss, "qnet", self.visualize_dir)

@@ -14,6 +14,7 @@ replay as preallocated flat device rings:
   machin_amd.ops.sumtree.DeviceSumTree — stratified sampling and
   priority updates never leave the GPU.
 """
+import threading
 from typing import Dict, Tuple, Union
 
 import torch as t
@@ -160,3 +161,251 @@ class DevicePrioritizedBuffer(DeviceReplayBuffer):
         super().clear()
         self.wt_tree.weights.zero_()
         self._max_priority = t.ones((), device=self.device)
+
+
+class DeviceTransitionBuffer:
+    """Algorithm-facing adapter over the HBM flat rings.
+
+    Speaks the same API as :class:`..buffers.buffer.Buffer` /
+    :class:`..buffers.prioritized_buffer.PrioritizedBuffer`
+    (``store_episode`` / ``sample_batch`` / ``update_priority`` /
+    ``size`` / ``clear``) so an algorithm constructed with
+    ``replay_device="cuda:0"`` transparently keeps its whole replay in
+    HBM: storing an episode is one H2D copy per attribute, sampling is
+    a device-side ``index_select`` per attribute, and PER priorities
+    live in a :class:`machin_amd.ops.sumtree.DeviceSumTree` — no
+    python-per-transition work and no host round trips on the update
+    path (reference stores python Transition objects per step,
+    machin/frame/buffers/storage.py:85-114).
+
+    The attribute layout is inferred from the first stored transition:
+    major attrs (state/action/next_state) become one ring per inner
+    key ("state/k"), scalar attrs and scalar custom attrs become
+    float32 rings, tensor custom attrs keep their shape/dtype.
+
+    Divergence from the reference ring: eviction is per-transition
+    (oldest first), not whole-episode.
+    """
+
+    accepts_tensor_priorities = True
+
+    _MAJOR = ("state", "action", "next_state")
+
+    def __init__(
+        self,
+        buffer_size: int,
+        device: Union[str, t.device] = "cuda:0",
+        prioritized: bool = False,
+        epsilon: float = 1e-2,
+        alpha: float = 0.6,
+        beta: float = 0.4,
+        beta_increment_per_sampling: float = 0.001,
+    ):
+        self.buffer_size = int(buffer_size)
+        self.buffer_device = t.device(device)
+        self.prioritized = prioritized
+        self._per_kwargs = dict(
+            epsilon=epsilon, alpha=alpha, beta=beta,
+            beta_increment_per_sampling=beta_increment_per_sampling,
+        )
+        self._inner = None
+        self._attr_order = None  # sample_attrs default order
+        self._pinned: Dict[str, t.Tensor] = {}
+        self._pinned_event = None
+        # concurrent RPC store services + the learner's own
+        # sample/update calls share this buffer (reference analog:
+        # wr_lock in machin/frame/buffers/buffer_d.py:77)
+        self._lock = threading.RLock()
+
+    # -- layout --------------------------------------------------------
+    @staticmethod
+    def _flatten(transition) -> Dict[str, t.Tensor]:
+        """One transition (Transition or dict) -> flat {col: tensor
+        with leading batch dim 1}."""
+        if not isinstance(transition, dict):
+            transition = {
+                k: getattr(transition, k) for k in transition.keys()
+            }
+        flat = {}
+        for attr, value in transition.items():
+            if attr in DeviceTransitionBuffer._MAJOR:
+                for k, v in value.items():
+                    flat[f"{attr}/{k}"] = v
+            elif t.is_tensor(value):
+                if value.dim() >= 2:
+                    flat[attr] = value  # already [1, ...]
+                elif value.dim() == 1:
+                    flat[attr] = value.view(1, -1)
+                else:
+                    flat[attr] = value.reshape(1)
+            else:
+                flat[attr] = t.tensor([float(value)], dtype=t.float32)
+        return flat
+
+    def _ensure_inner(self, flat: Dict[str, t.Tensor]):
+        if self._inner is not None:
+            return
+        spec = {
+            k: (tuple(v.shape[1:]), v.dtype) for k, v in flat.items()
+        }
+        if self.prioritized:
+            self._inner = DevicePrioritizedBuffer(
+                self.buffer_size, spec, self.buffer_device,
+                **self._per_kwargs,
+            )
+        else:
+            self._inner = DeviceReplayBuffer(
+                self.buffer_size, spec, self.buffer_device
+            )
+        self._attr_order = sorted(
+            {k.split("/", 1)[0] for k in spec}
+        )
+
+    # -- storing -------------------------------------------------------
+    def flatten_episode(self, episode, required_attrs=()) -> Dict[str, t.Tensor]:
+        """Episode -> one flat {col: [L, ...] tensor} dict (the wire
+        format used by the distributed device buffer)."""
+        if len(episode) == 0:
+            raise ValueError("Episode must be non-empty.")
+        flats = [self._flatten(tr) for tr in episode]
+        for attr in required_attrs:
+            if attr in self._MAJOR:
+                ok = any(k.startswith(attr + "/") for k in flats[0])
+            else:
+                ok = attr in flats[0]
+            if not ok:
+                raise ValueError(f"Transition missing attribute {attr!r}")
+        return {
+            k: t.cat([f[k] for f in flats], dim=0).detach()
+            for k in flats[0]
+        }
+
+    def store_flat(self, batch: Dict[str, t.Tensor]):
+        """Store a pre-flattened batch: stage host tensors through a
+        reusable pinned slab and issue ONE async H2D copy per
+        attribute (the BASELINE "rollouts pinned in host DRAM,
+        streamed into HBM" path)."""
+        with self._lock:
+            self._ensure_inner(batch)
+            self._inner.store_batch(self._stage(batch))
+            if self._pinned_event is not None:
+                self._pinned_event.record()
+
+    def store_episode(self, episode, required_attrs=("state", "action",
+                      "next_state", "reward", "terminal"), **__):
+        self.store_flat(self.flatten_episode(episode, required_attrs))
+
+    def _stage(self, batch: Dict[str, t.Tensor]) -> Dict[str, t.Tensor]:
+        if self.buffer_device.type != "cuda":
+            return batch
+        # pinned slabs are reused across calls: wait until the
+        # previous async H2D copies out of them have completed
+        if self._pinned_event is not None:
+            self._pinned_event.synchronize()
+        else:
+            self._pinned_event = t.cuda.Event()
+        out = {}
+        for k, v in batch.items():
+            if v.is_cuda:
+                out[k] = v
+                continue
+            n = v.shape[0]
+            slab = self._pinned.get(k)
+            if slab is None or slab.shape[0] < n \
+                    or slab.shape[1:] != v.shape[1:]:
+                slab = t.empty(
+                    (max(n, 64), *v.shape[1:]), dtype=v.dtype
+                ).pin_memory()
+                self._pinned[k] = slab
+            slab[:n].copy_(v)
+            out[k] = slab[:n]
+        return out
+
+    def append(self, transition, *_, **__):
+        self.store_episode([transition], required_attrs=())
+
+    store_transition = append
+
+    # -- sampling ------------------------------------------------------
+    def _structure(self, cols: Dict[str, t.Tensor], sample_attrs):
+        used = set()
+        out = []
+        for attr in sample_attrs:
+            if attr in self._MAJOR:
+                sub = {
+                    k.split("/", 1)[1]: v
+                    for k, v in cols.items()
+                    if k.startswith(attr + "/")
+                }
+                used.update(
+                    k for k in cols if k.startswith(attr + "/")
+                )
+                out.append(sub)
+            elif attr == "*":
+                rest = {
+                    k: v for k, v in cols.items()
+                    if k not in used and "/" not in k
+                }
+                out.append(rest)
+            else:
+                v = cols[attr]
+                if v.dim() == 1:
+                    v = v.view(-1, 1)
+                out.append(v)
+                used.add(attr)
+        return tuple(out)
+
+    def sample_batch(
+        self,
+        batch_size: int,
+        concatenate: bool = True,
+        device=None,
+        sample_method=None,
+        sample_attrs=None,
+        additional_concat_custom_attrs=None,
+        *_,
+        **__,
+    ):
+        if not concatenate:
+            raise ValueError(
+                "DeviceTransitionBuffer only supports concatenated "
+                "sampling (flat device rings)."
+            )
+        with self._lock:
+            empty = self._inner is None or self._inner.size() == 0 \
+                or batch_size <= 0
+            if self.prioritized:
+                if empty:
+                    return 0, None, None, None
+                cols, idx, is_weight = self._inner.sample_batch(batch_size)
+                attrs = sample_attrs or (self._attr_order + ["*"])
+                return (
+                    batch_size,
+                    self._structure(cols, attrs),
+                    idx,
+                    is_weight,
+                )
+            if empty:
+                return 0, None
+            cols = self._inner.sample_batch(batch_size)
+            attrs = sample_attrs or (self._attr_order + ["*"])
+            return batch_size, self._structure(cols, attrs)
+
+    # -- PER interface -------------------------------------------------
+    def update_priority(self, priorities, indexes):
+        priorities = t.as_tensor(priorities, dtype=t.float32)
+        indexes = t.as_tensor(indexes)
+        with self._lock:
+            self._inner.update_priority(priorities, indexes)
+
+    # -- bookkeeping ---------------------------------------------------
+    def size(self) -> int:
+        return 0 if self._inner is None else self._inner.size()
+
+    def __len__(self):
+        return self.size()
+
+    def clear(self):
+        with self._lock:
+            if self._inner is not None:
+                self._inner.clear()

@@ -30,7 +30,11 @@ class DeviceSumTree:
         self.weights = t.zeros(
             2 * self.capacity, dtype=t.float32, device=self.device
         )
-        self._ext = _require_ext()
+        # CUDA tensors REQUIRE the gfx950 extension (no silent eager
+        # fallback on a GPU box); a CPU-device tree runs the torch
+        # fallback below so the device data path is testable in
+        # CPU-only CI (gloo multi-process tests).
+        self._ext = _require_ext() if self.device.type == "cuda" else None
 
     # -- accessors -----------------------------------------------------
     def get_weight_sum(self) -> float:
@@ -56,21 +60,46 @@ class DeviceSumTree:
     def update_leaf_batch(self, weights: t.Tensor, indexes: t.Tensor):
         weights = weights.to(device=self.device, dtype=t.float32).contiguous()
         indexes = indexes.to(device=self.device, dtype=t.long).contiguous()
+        if self._ext is None:
+            self.weights[self.capacity + indexes] = weights
+            self._rebuild_cpu()
+            return
         self._ext.sumtree_update(
             self.weights, indexes, weights, self.capacity, self.depth
         )
+
+    def _rebuild_cpu(self):
+        """Full bottom-up rebuild (CPU fallback path)."""
+        base, size = self.capacity, self.capacity
+        while size > 1:
+            parents = self.weights[base : base + size].view(-1, 2).sum(1)
+            self.weights[base // 2 : base // 2 + size // 2] = parents
+            base //= 2
+            size //= 2
 
     def update_all_leaves(self, weights: t.Tensor):
         if weights.numel() != self.size:
             raise ValueError(f"Expected {self.size} weights.")
         leaves = self.weights[self.capacity : self.capacity + self.size]
         leaves.copy_(weights.to(device=self.device, dtype=t.float32))
+        if self._ext is None:
+            self._rebuild_cpu()
+            return
         self._ext.sumtree_build(self.weights, self.capacity)
 
     # -- queries -------------------------------------------------------
     def find_leaf_index(self, prefix_weights: t.Tensor) -> t.Tensor:
         """Map prefix weights in [0, sum) to leaf indexes."""
         u = prefix_weights.to(device=self.device, dtype=t.float32).contiguous()
+        if self._ext is None:
+            u = u.clone()
+            idx = t.ones_like(u, dtype=t.long)
+            for _ in range(self.depth):
+                left = self.weights[2 * idx]
+                right = (u >= left) & (left < self.weights[idx])
+                u = u - left * right
+                idx = 2 * idx + right.long()
+            return (idx - self.capacity).clamp_(0, self.size - 1)
         return self._ext.sumtree_sample(
             self.weights, u, self.capacity, self.depth, self.size
         )

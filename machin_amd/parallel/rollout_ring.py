@@ -130,3 +130,111 @@ class RolloutRing:
             for k, buf in self.data.items():
                 pool[k][pos].copy_(buf[slot_id],
                                    non_blocking=non_blocking)
+
+
+class EpisodeSegmentCodec:
+    """Maps algorithm-level episodes (lists of Transition/dicts) onto
+    fixed-shape ring slots so IMPALA actors can store rollouts through
+    shared memory instead of control-plane RPC.
+
+    Slot layout (unroll T):
+      * ``state/<k>``  [T, ...]   episode states (zero-padded)
+      * ``action/<k>`` [T, ...]   taken actions
+      * ``reward``     [T]        float32
+      * ``terminal``   [T]        float32
+      * ``behavior_logp`` [T]     float32 (required by V-trace)
+      * ``boot_state/<k>`` [...]  next_state of the segment's last
+                                  transition (bootstrap input)
+      * ``length``     []         int64 valid steps
+
+    Episodes longer than T are split into consecutive segments, each
+    bootstrapping from its own tail — the standard IMPALA unroll.
+    """
+
+    def __init__(self, unroll: int):
+        self.unroll = int(unroll)
+
+    # -- spec ----------------------------------------------------------
+    def spec_from_episode(self, episode) -> Dict[str, Tuple[Tuple[int, ...], t.dtype]]:
+        tr = episode[0]
+        if not isinstance(tr, dict):
+            tr = {k: getattr(tr, k) for k in tr.keys()}
+        T = self.unroll
+        spec = {
+            "reward": ((T,), t.float32),
+            "terminal": ((T,), t.float32),
+            "behavior_logp": ((T,), t.float32),
+            "length": ((), t.long),
+        }
+        for k, v in tr["state"].items():
+            spec[f"state/{k}"] = ((T, *v.shape[1:]), v.dtype)
+        for k, v in tr["action"].items():
+            spec[f"action/{k}"] = ((T, *v.shape[1:]), v.dtype)
+        for k, v in tr["next_state"].items():
+            spec[f"boot_state/{k}"] = (tuple(v.shape[1:]), v.dtype)
+        return spec
+
+    # -- actor side ----------------------------------------------------
+    @staticmethod
+    def _scalar(v):
+        return float(v.reshape(-1)[0]) if t.is_tensor(v) else float(v)
+
+    def write_episode(self, ring: "RolloutRing", episode,
+                      timeout: float = None):
+        """Split an episode into segments and write each into a free
+        slot. Blocks when the ring is full (learner backpressure)."""
+        T = self.unroll
+        dicts = []
+        for tr in episode:
+            if not isinstance(tr, dict):
+                tr = {k: getattr(tr, k) for k in tr.keys()}
+            if "action_log_prob" not in tr:
+                raise ValueError(
+                    "IMPALA ring transitions require 'action_log_prob'."
+                )
+            dicts.append(tr)
+        for start in range(0, len(dicts), T):
+            chunk = dicts[start : start + T]
+            L = len(chunk)
+            slot_id = ring.acquire(timeout=timeout)
+            slot = ring.slot(slot_id)
+            for k in chunk[0]["state"]:
+                col = slot[f"state/{k}"]
+                col[:L] = t.cat([c["state"][k] for c in chunk], dim=0)
+                if L < T:
+                    col[L:] = 0
+            for k in chunk[0]["action"]:
+                col = slot[f"action/{k}"]
+                col[:L] = t.cat([c["action"][k] for c in chunk], dim=0)
+                if L < T:
+                    col[L:] = 0
+            for k in chunk[-1]["next_state"]:
+                slot[f"boot_state/{k}"].copy_(
+                    chunk[-1]["next_state"][k][0]
+                )
+            slot["reward"][:L] = t.as_tensor(
+                [self._scalar(c["reward"]) for c in chunk]
+            )
+            slot["terminal"][:L] = t.as_tensor(
+                [self._scalar(c["terminal"]) for c in chunk]
+            )
+            slot["behavior_logp"][:L] = t.as_tensor(
+                [self._scalar(c["action_log_prob"]) for c in chunk]
+            )
+            if L < T:
+                slot["reward"][L:] = 0
+                slot["terminal"][L:] = 1.0
+                slot["behavior_logp"][L:] = 0
+            slot["length"].fill_(L)
+            ring.commit(slot_id)
+
+
+def make_episode_ring(sample_episode, unroll: int, slots: int,
+                      ctx=None) -> Tuple["RolloutRing", EpisodeSegmentCodec]:
+    """Build a RolloutRing sized for the given episode shape."""
+    codec = EpisodeSegmentCodec(unroll)
+    ring = RolloutRing(
+        slots=slots, spec=codec.spec_from_episode(sample_episode),
+        ctx=ctx,
+    )
+    return ring, codec

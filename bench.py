@@ -1,11 +1,18 @@
-"""Flagship benchmark: IMPALA Atari-CNN learner throughput on MI355X.
+"""Flagship benchmark: IMPALA Atari-CNN end-to-end throughput on
+MI355X — BOTH halves of the BASELINE.json headline metric:
 
-Measures the BASELINE.json headline metric — learner samples/sec for
-IMPALA with the Nature CNN on Atari-shaped synthetic data — with the
-full learner step in the timed region: uint8 frame normalization,
-bf16 forward (MIOpen convs / hipBLASLt linears), V-trace targets (gfx950
-HIP kernel), policy/value/entropy losses, backward, bucketed RCCL
-all-reduce over xGMI (N>1), gradient clip and optimizer step.
+* ``learner samples/sec`` — samples consumed by gradient updates
+  (u8 dequant, bf16 forward, V-trace gfx950 kernel, losses, backward,
+  bucketed RCCL all-reduce over xGMI for N>1, clip, optimizer step);
+* ``env-steps/sec`` — fresh environment steps ingested while training
+  runs: a host actor farm writes rollout segments into a
+  shared-memory RolloutRing; a background thread drains them through
+  pinned staging (hipHostRegister direct DMA when available) onto a
+  side stream into an HBM-resident segment pool the learner trains
+  from.
+
+Default mode is the full pipeline; ``--learner-only`` measures the
+bare learner loop from GPU-resident pools (round-1 configuration).
 
 Usage:
     python bench.py [--gpus N] [--steps K] [--warmup W]
@@ -17,11 +24,185 @@ Scaling is WEAK: per-GPU work (unroll x env_batch) is fixed as N grows.
 import argparse
 import json
 import os
+import threading
 import time
 
 import torch as t
 import torch.distributed as dist
 import torch.nn as nn
+
+
+def _actor_loop(ring, actor_id, stop_flag, steps_counter, unroll,
+                frames, actions):
+    """Synthetic Atari actor process: fills rollout segments in shared
+    memory. A real actor would run ALE + a policy net (no ROMs without
+    network access); a pre-generated segment library is cycled with
+    cheap mutation so the measured cost is the TRANSPORT pipeline
+    (shared-memory writes, slot bookkeeping), not RNG."""
+    t.set_num_threads(1)
+    try:
+        os.nice(5)  # the learner's host threads have priority
+    except OSError:
+        pass
+    gen = t.Generator().manual_seed(actor_id)
+    library = [
+        {
+            "frames": t.randint(0, 256, (unroll, frames, 84, 84),
+                                dtype=t.uint8, generator=gen),
+            "actions": t.randint(0, actions, (unroll,), generator=gen),
+            "behavior_logp": -t.rand(unroll, generator=gen) * 2.0,
+            "rewards": t.rand(unroll, generator=gen),
+            "terminals": t.zeros(unroll),
+        }
+        for _ in range(8)
+    ]
+    i = 0
+    while not stop_flag[0]:
+        try:
+            slot_id = ring.acquire(timeout=1.0)
+        except Exception:  # noqa: BLE001 - queue.Empty on shutdown
+            continue
+        src = library[i % len(library)]
+        i += 1
+        slot = ring.slot(slot_id)
+        slot["frames"].copy_(src["frames"])
+        slot["frames"][0, 0, 0, 0] = i % 256  # per-segment variation
+        slot["actions"].copy_(src["actions"])
+        slot["behavior_logp"].copy_(src["behavior_logp"])
+        slot["rewards"].copy_(src["rewards"])
+        slot["terminals"].copy_(src["terminals"])
+        ring.commit(slot_id)
+        with steps_counter.get_lock():
+            steps_counter.value += unroll
+
+
+class RingPipeline:
+    """Actor farm -> shared-memory ring -> pinned staging -> HBM pool
+    (the BASELINE "async rollout staging" path), feeding the learner
+    with t-major batches gathered on device."""
+
+    def __init__(self, device, unroll, env_batch, actions, frames=4,
+                 actors=6, slots=256, pool_segments=4096):
+        import torch.multiprocessing as mp
+
+        from machin_amd.parallel.rollout_ring import RolloutRing
+
+        self.device = device
+        self.unroll = unroll
+        self.env_batch = env_batch
+        ctx = mp.get_context("spawn")
+        self.spec = {
+            "frames": ((unroll, frames, 84, 84), t.uint8),
+            "actions": ((unroll,), t.long),
+            "behavior_logp": ((unroll,), t.float32),
+            "rewards": ((unroll,), t.float32),
+            "terminals": ((unroll,), t.float32),
+        }
+        self.ring = RolloutRing(slots=slots, spec=self.spec, ctx=ctx)
+        self.stop_flag = t.zeros(1, dtype=t.uint8).share_memory_()
+        self.steps_counter = ctx.Value("q", 0)
+        self.actors = [
+            ctx.Process(
+                target=_actor_loop,
+                args=(self.ring, a, self.stop_flag, self.steps_counter,
+                      unroll, frames, actions),
+                daemon=True,
+            )
+            for a in range(actors)
+        ]
+        M = pool_segments
+        self.M = M
+        self.pool = {
+            k: t.zeros((M, *shape), dtype=dtype, device=device)
+            for k, (shape, dtype) in self.spec.items()
+        }
+        self.pinned = self.ring.make_pinned_staging(64)
+        self.registered = False
+        self.side = t.cuda.Stream()
+        self.filled = 0
+        self.write_pos = 0
+        self._stop_ingest = False
+        self._ingester = None
+
+    def start(self):
+        for p in self.actors:
+            p.start()
+        self.registered = self.ring.host_register()
+        # pre-fill the pool so the first batches sample real segments
+        while self.filled < min(self.env_batch, self.M):
+            self.ingest_once(timeout=2.0)
+        self._ingester = threading.Thread(
+            target=self._ingest_loop, daemon=True
+        )
+        self._ingester.start()
+
+    def ingest_once(self, timeout=0.05):
+        idx = self.ring.drain(max_slots=64, timeout=timeout)
+        if not idx:
+            return 0
+        wp, M, n = self.write_pos, self.M, len(idx)
+        with t.cuda.stream(self.side):
+            if self.registered:
+                positions = [(wp + i) % M for i in range(n)]
+                self.ring.upload_slots(idx, self.pool, positions)
+            else:
+                batch = self.ring.gather(
+                    idx, self.device, pinned=self.pinned
+                )
+                for k, v in batch.items():
+                    end = wp + n
+                    if end <= M:
+                        self.pool[k][wp:end] = v
+                    else:
+                        split = M - wp
+                        self.pool[k][wp:] = v[:split]
+                        self.pool[k][: end % M] = v[split:]
+        self.side.synchronize()
+        self.ring.release(idx)
+        self.write_pos = (wp + n) % M
+        self.filled = min(self.filled + n, M)
+        return n
+
+    def _ingest_loop(self):
+        # background staging, overlapped with training (host memcpys
+        # release the GIL)
+        while not self._stop_ingest:
+            self.ingest_once()
+
+    def sample(self):
+        """Gather a t-major training batch from the HBM pool."""
+        B, T = self.env_batch, self.unroll
+        seg = t.randint(0, max(self.filled, 1), (B,), device=self.device)
+        p = self.pool
+        return {
+            "frames": p["frames"].index_select(0, seg)
+            .permute(1, 0, 2, 3, 4).reshape(T * B, *p["frames"].shape[2:])
+            .to(memory_format=t.channels_last),
+            "actions": p["actions"].index_select(0, seg).t().contiguous(),
+            "behavior_logp": p["behavior_logp"].index_select(0, seg)
+            .t().contiguous(),
+            "rewards": p["rewards"].index_select(0, seg).t().contiguous(),
+            "terminals": p["terminals"].index_select(0, seg)
+            .t().contiguous(),
+        }
+
+    def reset_env_steps(self):
+        with self.steps_counter.get_lock():
+            self.steps_counter.value = 0
+
+    def env_steps(self) -> int:
+        with self.steps_counter.get_lock():
+            return self.steps_counter.value
+
+    def stop(self):
+        self._stop_ingest = True
+        if self._ingester is not None:
+            self._ingester.join(timeout=5)
+        self.stop_flag[0] = 1
+        for p in self.actors:
+            p.join(timeout=5)
+            if p.is_alive():
+                p.terminate()
 
 
 class ImpalaLearnerBench:
@@ -44,6 +225,9 @@ class ImpalaLearnerBench:
         grad_clip: float = 40.0,
         capturable: bool = False,
         fused_stem: bool = True,
+        bucket_mb: float = 32.0,
+        ddp_reduction: str = "all_reduce",
+        pipeline: RingPipeline = None,
     ):
         from machin_amd.model.nets.nature_cnn import (
             ActorCriticCNN,
@@ -65,6 +249,7 @@ class ImpalaLearnerBench:
         self.value_weight = value_weight
         self.discount = discount
         self.grad_clip = grad_clip
+        self.pipeline = pipeline
 
         self.fused_stem = fused_stem and dtype == t.bfloat16 and frames == 4
         if self.fused_stem:
@@ -76,7 +261,10 @@ class ImpalaLearnerBench:
         if distributed:
             from machin_amd.parallel.ddp import GradReducer
 
-            self.reducer = GradReducer(self.model, bucket_cap_mb=32.0)
+            self.reducer = GradReducer(
+                self.model, bucket_cap_mb=bucket_mb,
+                reduction=ddp_reduction,
+            )
             with t.no_grad():
                 for p in self.model.parameters():
                     dist.broadcast(p.data, src=0)
@@ -85,35 +273,38 @@ class ImpalaLearnerBench:
             capturable=capturable,
         )
 
-        # synthetic rollout pool (uint8 frames like a real Atari actor
-        # feed; behavior log-probs from a slightly-off policy)
-        TB = unroll * env_batch
-        # per-rank data seeds: ranks must NOT train on identical data,
-        # or the all-reduce degenerates to a no-op check
-        rank = int(os.environ.get("RANK", "0"))
-        g = t.Generator(device="cpu").manual_seed(1234 + rank * 1000)
+        # synthetic rollout pool for --learner-only mode (uint8 frames
+        # like a real Atari actor feed)
         self.pool = []
-        for _ in range(pool_size):
-            self.pool.append(
-                {
-                    "frames": t.randint(
-                        0, 256, (TB, frames, 84, 84), dtype=t.uint8,
-                        generator=g,
-                    ).to(self.device).to(memory_format=t.channels_last),
-                    "actions": t.randint(
-                        0, action_num, (unroll, env_batch), generator=g
-                    ).to(self.device),
-                    "behavior_logp": (
-                        -t.rand(unroll, env_batch, generator=g) * 2.0
-                    ).to(self.device),
-                    "rewards": t.rand(unroll, env_batch, generator=g).to(
-                        self.device
-                    ),
-                    "terminals": (
-                        t.rand(unroll, env_batch, generator=g) > 0.98
-                    ).float().to(self.device),
-                }
-            )
+        if pipeline is None:
+            TB = unroll * env_batch
+            # per-rank data seeds: ranks must NOT train on identical
+            # data, or the all-reduce degenerates to a no-op check
+            rank = int(os.environ.get("RANK", "0"))
+            g = t.Generator(device="cpu").manual_seed(1234 + rank * 1000)
+            for _ in range(pool_size):
+                self.pool.append(
+                    {
+                        "frames": t.randint(
+                            0, 256, (TB, frames, 84, 84), dtype=t.uint8,
+                            generator=g,
+                        ).to(self.device)
+                        .to(memory_format=t.channels_last),
+                        "actions": t.randint(
+                            0, action_num, (unroll, env_batch),
+                            generator=g,
+                        ).to(self.device),
+                        "behavior_logp": (
+                            -t.rand(unroll, env_batch, generator=g) * 2.0
+                        ).to(self.device),
+                        "rewards": t.rand(
+                            unroll, env_batch, generator=g
+                        ).to(self.device),
+                        "terminals": (
+                            t.rand(unroll, env_batch, generator=g) > 0.98
+                        ).float().to(self.device),
+                    }
+                )
         self._pool_i = 0
         self._graph = None
         self._static_in = None
@@ -122,6 +313,8 @@ class ImpalaLearnerBench:
     def step(self):
         """One learner step; returns the detached loss TENSOR (no
         host synchronization on the hot path)."""
+        if self.pipeline is not None:
+            return self._step_body(self.pipeline.sample())
         data = self.pool[self._pool_i]
         self._pool_i = (self._pool_i + 1) % len(self.pool)
         if self._graph is not None:
@@ -190,9 +383,11 @@ class ImpalaLearnerBench:
         """Capture the whole learner step in a hipGraph: the Nature
         CNN is small, so kernel-launch overhead is a real cost at low
         batch — one graph replay replaces ~300 launches. Single-GPU
-        only (RCCL collectives stay outside graphs here)."""
+        learner-only mode (RCCL collectives stay outside graphs)."""
         if self.reducer is not None:
             raise RuntimeError("graph capture is single-GPU only")
+        if self.pipeline is not None:
+            raise RuntimeError("graph capture requires --learner-only")
         data = self.pool[0]
         self._static_in = {
             k: v.clone() for k, v in data.items()
@@ -211,17 +406,27 @@ class ImpalaLearnerBench:
 def main():
     parser = argparse.ArgumentParser()
     parser.add_argument("--gpus", type=int, default=1)
-    parser.add_argument("--steps", type=int, default=30)
-    parser.add_argument("--warmup", type=int, default=10)
+    parser.add_argument("--steps", type=int, default=150)
+    parser.add_argument("--warmup", type=int, default=30)
     parser.add_argument("--unroll", type=int, default=20)
     parser.add_argument("--env-batch", type=int, default=2048)
     parser.add_argument("--actions", type=int, default=6)
+    parser.add_argument("--actors", type=int, default=0,
+                        help="actor processes per rank (0 = auto)")
+    parser.add_argument("--slots", type=int, default=256)
+    parser.add_argument("--pool-segments", type=int, default=4096)
+    parser.add_argument("--learner-only", action="store_true",
+                        help="skip the actor farm/ring; measure the "
+                             "bare learner loop from resident pools")
     parser.add_argument("--graph", action="store_true",
-                        help="capture the learner step in a hipGraph")
+                        help="capture the learner step in a hipGraph "
+                             "(learner-only mode)")
+    parser.add_argument("--bucket-mb", type=float, default=32.0,
+                        help="GradReducer bucket size (MiB)")
+    parser.add_argument("--ddp-reduction", default="all_reduce",
+                        choices=["all_reduce", "reduce_scatter"])
     parser.add_argument("--no-fused-stem", action="store_true",
-                        help="fall back to MIOpen for the stem conv "
-                             "(the fused u8 kernels measure +12%% "
-                             "end-to-end: 3.81 vs 3.41 M samples/s)")
+                        help="fall back to MIOpen for the stem conv")
     args = parser.parse_args()
 
     world_size = int(os.environ.get("WORLD_SIZE", "1"))
@@ -246,6 +451,21 @@ def main():
     t.manual_seed(42 + rank)
     t.backends.cudnn.benchmark = True
 
+    if args.graph and not args.learner_only:
+        args.learner_only = True
+
+    pipeline = None
+    if not args.learner_only:
+        n_actors = args.actors
+        if n_actors <= 0:
+            cpus = os.cpu_count() or 16
+            n_actors = max(2, min(8, cpus // (2 * world_size)))
+        pipeline = RingPipeline(
+            device, args.unroll, args.env_batch, args.actions,
+            actors=n_actors, slots=args.slots,
+            pool_segments=args.pool_segments,
+        )
+
     bench = ImpalaLearnerBench(
         device=device,
         unroll=args.unroll,
@@ -254,8 +474,13 @@ def main():
         distributed=distributed,
         capturable=args.graph and not distributed,
         fused_stem=not args.no_fused_stem,
+        bucket_mb=args.bucket_mb,
+        ddp_reduction=args.ddp_reduction,
+        pipeline=pipeline,
     )
 
+    if pipeline is not None:
+        pipeline.start()
     if args.graph and not distributed:
         bench.capture_graph()
     for _ in range(args.warmup):
@@ -263,17 +488,25 @@ def main():
     if distributed:
         dist.barrier()
     t.cuda.synchronize()
+    if pipeline is not None:
+        pipeline.reset_env_steps()
 
     t0 = time.perf_counter()
     for _ in range(args.steps):
         bench.step()
     t.cuda.synchronize()
     elapsed = time.perf_counter() - t0
+    env_steps = pipeline.env_steps() if pipeline is not None else 0
     if distributed:
         et = t.tensor([elapsed], device=device)
         dist.all_reduce(et, op=dist.ReduceOp.MAX)
+        es = t.tensor([float(env_steps)], device=device)
+        dist.all_reduce(es, op=dist.ReduceOp.SUM)
         dist.barrier()
         elapsed = float(et.item())
+        env_steps = int(es.item())
+    if pipeline is not None:
+        pipeline.stop()
 
     samples_per_step = args.unroll * args.env_batch
     value = samples_per_step * args.steps * n_gpus / elapsed
@@ -295,6 +528,14 @@ def main():
                     "vs_baseline": None,
                     "dtype": "bf16",
                     "data": "synthetic",
+                    "env_steps_per_sec": (
+                        env_steps / elapsed if pipeline is not None
+                        else None
+                    ),
+                    "pipeline": (
+                        "learner_only" if pipeline is None
+                        else "actors+ring+hbm_pool"
+                    ),
                     "config": {
                         "model": "impala_nature_cnn_atari",
                         "global_batch": samples_per_step * n_gpus,

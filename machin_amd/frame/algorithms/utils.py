@@ -20,18 +20,39 @@ import torch.nn as nn
 # target-network updates
 # ----------------------------------------------------------------------
 def soft_update(target_net: nn.Module, source_net: nn.Module, update_rate: float):
-    """Polyak: target = target·(1−τ) + source·τ over every parameter."""
+    """Polyak: target = target·(1−τ) + source·τ over every parameter.
+
+    On GPU a :class:`machin_amd.ops.FusedPolyak` plan is cached on the
+    target net: one kernel launch per call with a precomputed device
+    pointer table (beats both the per-param reference loop,
+    machin/frame/algorithms/utils.py:8-27, and per-call table builds).
+    Set MACHIN_AMD_FUSED_POLYAK=0 to force torch _foreach.
+    """
+    import os
+
     tgt = [p.data for p in target_net.parameters()]
     src = [p.data for p in source_net.parameters()]
     if not tgt:
         return
-    if tgt[0].is_cuda:
-        from ...ops import polyak_update_
+    if (
+        tgt[0].is_cuda
+        and os.environ.get("MACHIN_AMD_FUSED_POLYAK", "1") != "0"
+    ):
+        from ... import ops
 
-        polyak_update_(tgt, src, update_rate)
-    else:
-        t._foreach_mul_(tgt, 1.0 - update_rate)
-        t._foreach_add_(tgt, src, alpha=update_rate)
+        if ops.available():
+            plan = target_net.__dict__.get("_machin_polyak_plan")
+            if plan is None or not plan.matches(tgt, src):
+                try:
+                    plan = ops.FusedPolyak(tgt, src)
+                    target_net.__dict__["_machin_polyak_plan"] = plan
+                except ValueError:
+                    plan = None
+            if plan is not None:
+                plan(update_rate)
+                return
+    t._foreach_mul_(tgt, 1.0 - update_rate)
+    t._foreach_add_(tgt, src, alpha=update_rate)
 
 
 def hard_update(target_net: nn.Module, source_net: nn.Module):

@@ -25,6 +25,61 @@ def _use_fused_polyak() -> bool:
     return os.environ.get("MACHIN_AMD_FUSED_POLYAK", "0") == "1"
 
 
+class FusedPolyak:
+    """Precomputed single-launch polyak update over a parameter list.
+
+    The round-1 kernel lost to torch ``_foreach`` (0.106 vs 0.017 ms,
+    profiles/kernel_bench_r01.json) because every call rebuilt the
+    pointer table on the host and shipped it H2D. This plan builds the
+    device table ONCE; each call is one kernel launch with zero host
+    setup. Construction raises ValueError if any tensor pair is not
+    eligible (contiguous fp32 CUDA) — callers fall back to _foreach.
+    """
+
+    def __init__(self, targets: List[t.Tensor], sources: List[t.Tensor]):
+        ext = _require_ext()
+        n = len(targets)
+        if n == 0:
+            raise ValueError("empty tensor list")
+        for tt, ss in zip(targets, sources):
+            if not (
+                tt.is_cuda and ss.is_cuda
+                and tt.is_contiguous() and ss.is_contiguous()
+                and tt.dtype == t.float32 and ss.dtype == t.float32
+                and tt.numel() == ss.numel()
+            ):
+                raise ValueError("ineligible tensor pair for fused polyak")
+        table = t.empty(3 * n + 1, dtype=t.int64)
+        total = 0
+        for i, (tt, ss) in enumerate(zip(targets, sources)):
+            table[2 * i] = tt.data_ptr()
+            table[2 * i + 1] = ss.data_ptr()
+            table[2 * n + i] = total
+            total += tt.numel()
+        table[3 * n] = total
+        self.table = table.to(targets[0].device)
+        self.n = n
+        self.total = total
+        self._sig = tuple(int(x.data_ptr()) for x in targets) + tuple(
+            int(x.data_ptr()) for x in sources
+        )
+        self._ext = ext
+
+    def matches(self, targets, sources) -> bool:
+        """Cheap revalidation: storage pointers unchanged."""
+        if len(targets) != self.n:
+            return False
+        sig = tuple(int(x.data_ptr()) for x in targets) + tuple(
+            int(x.data_ptr()) for x in sources
+        )
+        return sig == self._sig
+
+    def __call__(self, tau: float):
+        self._ext.multi_tensor_polyak_cached(
+            self.table, self.n, self.total, float(tau)
+        )
+
+
 def _load_ext():
     global _ext, _ext_error
     if _ext is not None or _ext_error is not None:

@@ -1,7 +1,10 @@
-// Python bindings for the machin_amd gfx950 kernels.
+// Python bindings for the machin_amd gfx950 kernels — written
+// directly against PyTorch-ROCm's native HIP API (the
+// "masquerading-as-CUDA" device layer every ROCm extension runs on),
+// so the tree carries NO hipify-generated sources.
 #include <torch/extension.h>
-#include <ATen/cuda/CUDAContext.h>
-#include <c10/cuda/CUDAGuard.h>
+#include <ATen/hip/HIPContext.h>
+#include <ATen/hip/impl/HIPGuardImplMasqueradingAsCUDA.h>
 
 #include <cstdint>
 #include <vector>
@@ -21,6 +24,8 @@ void gae_launch(const float*, const float*, const float*, const float*,
 void vtrace_launch(const float*, const float*, const float*, const float*,
                    const float*, const float*, float*, float*, int64_t,
                    int64_t, float, float, float, float, hipStream_t);
+void nstep_returns_launch(const float*, const float*, float*, int64_t,
+                          int64_t, float, int, hipStream_t);
 void categorical_projection_launch(const float*, const float*, const float*,
                                    float*, int64_t, int64_t, float, float,
                                    float, hipStream_t);
@@ -53,7 +58,7 @@ void check_f32_cuda(const Tensor& t, const char* name) {
 }
 
 hipStream_t current_stream() {
-  return at::cuda::getCurrentCUDAStream().stream();
+  return at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
 }
 
 // ------------------------------------------------------------------
@@ -66,7 +71,7 @@ void sumtree_update(Tensor tree, Tensor idx, Tensor w, int64_t capacity,
   TORCH_CHECK(idx.is_cuda() && idx.scalar_type() == at::kLong &&
                   idx.is_contiguous(),
               "idx must be contiguous int64 CUDA");
-  const at::cuda::OptionalCUDAGuard guard(tree.device());
+  const at::hip::OptionalHIPGuardMasqueradingAsCUDA guard(tree.device());
   sumtree_update_launch(tree.data_ptr<float>(), idx.data_ptr<int64_t>(),
                         w.data_ptr<float>(), idx.numel(), capacity,
                         (int)depth, current_stream());
@@ -74,7 +79,7 @@ void sumtree_update(Tensor tree, Tensor idx, Tensor w, int64_t capacity,
 
 void sumtree_build(Tensor tree, int64_t capacity) {
   check_f32_cuda(tree, "tree");
-  const at::cuda::OptionalCUDAGuard guard(tree.device());
+  const at::hip::OptionalHIPGuardMasqueradingAsCUDA guard(tree.device());
   sumtree_build_launch(tree.data_ptr<float>(), capacity, current_stream());
 }
 
@@ -82,7 +87,7 @@ Tensor sumtree_sample(Tensor tree, Tensor u, int64_t capacity, int64_t depth,
                       int64_t size) {
   check_f32_cuda(tree, "tree");
   check_f32_cuda(u, "u");
-  const at::cuda::OptionalCUDAGuard guard(tree.device());
+  const at::hip::OptionalHIPGuardMasqueradingAsCUDA guard(tree.device());
   Tensor out = at::empty({u.numel()}, u.options().dtype(at::kLong));
   sumtree_sample_launch(tree.data_ptr<float>(), u.data_ptr<float>(),
                         out.data_ptr<int64_t>(), u.numel(), capacity,
@@ -96,7 +101,7 @@ Tensor sumtree_sample(Tensor tree, Tensor u, int64_t capacity, int64_t depth,
 Tensor discounted_returns(Tensor rew, Tensor nd, Tensor bootstrap,
                           double gamma) {
   check_f32_cuda(rew, "rewards");
-  const at::cuda::OptionalCUDAGuard guard(rew.device());
+  const at::hip::OptionalHIPGuardMasqueradingAsCUDA guard(rew.device());
   int64_t T = rew.size(0), B = rew.size(1);
   Tensor out = at::empty_like(rew);
   discounted_returns_launch(rew.data_ptr<float>(), nd.data_ptr<float>(),
@@ -109,7 +114,7 @@ Tensor discounted_returns(Tensor rew, Tensor nd, Tensor bootstrap,
 Tensor gae(Tensor rew, Tensor val, Tensor next_val, Tensor nd, double gamma,
            double lam) {
   check_f32_cuda(rew, "rewards");
-  const at::cuda::OptionalCUDAGuard guard(rew.device());
+  const at::hip::OptionalHIPGuardMasqueradingAsCUDA guard(rew.device());
   int64_t T = rew.size(0), B = rew.size(1);
   Tensor out = at::empty_like(rew);
   gae_launch(rew.data_ptr<float>(), val.data_ptr<float>(),
@@ -124,7 +129,7 @@ std::vector<Tensor> vtrace(Tensor blp, Tensor tlp, Tensor rew, Tensor val,
                            double rho_clip, double c_clip,
                            double pg_rho_clip) {
   check_f32_cuda(rew, "rewards");
-  const at::cuda::OptionalCUDAGuard guard(rew.device());
+  const at::hip::OptionalHIPGuardMasqueradingAsCUDA guard(rew.device());
   int64_t T = rew.size(0), B = rew.size(1);
   Tensor vs = at::empty_like(rew);
   Tensor pg_adv = at::empty_like(rew);
@@ -137,13 +142,25 @@ std::vector<Tensor> vtrace(Tensor blp, Tensor tlp, Tensor rew, Tensor val,
   return {vs, pg_adv};
 }
 
+Tensor nstep_returns(Tensor rew, Tensor alive, double gamma, int64_t n) {
+  check_f32_cuda(rew, "rewards");
+  check_f32_cuda(alive, "alive");
+  const at::hip::OptionalHIPGuardMasqueradingAsCUDA guard(rew.device());
+  int64_t T = rew.size(0), B = rew.size(1);
+  Tensor out = at::empty_like(rew);
+  nstep_returns_launch(rew.data_ptr<float>(), alive.data_ptr<float>(),
+                       out.data_ptr<float>(), T, B, (float)gamma, (int)n,
+                       current_stream());
+  return out;
+}
+
 // ------------------------------------------------------------------
 // categorical projection
 // ------------------------------------------------------------------
 Tensor categorical_projection(Tensor next_dist, Tensor rew, Tensor nd,
                               double gamma, double v_min, double v_max) {
   check_f32_cuda(next_dist, "next_dist");
-  const at::cuda::OptionalCUDAGuard guard(next_dist.device());
+  const at::hip::OptionalHIPGuardMasqueradingAsCUDA guard(next_dist.device());
   int64_t B = next_dist.size(0), A = next_dist.size(1);
   TORCH_CHECK(A <= 256, "categorical projection supports at most 256 atoms");
   Tensor out = at::empty_like(next_dist);
@@ -162,7 +179,7 @@ void multi_tensor_polyak(std::vector<Tensor> targets,
                          std::vector<Tensor> sources, double tau) {
   TORCH_CHECK(targets.size() == sources.size(), "list size mismatch");
   if (targets.empty()) return;
-  const at::cuda::OptionalCUDAGuard guard(targets[0].device());
+  const at::hip::OptionalHIPGuardMasqueradingAsCUDA guard(targets[0].device());
   int64_t n = (int64_t)targets.size();
   // host-side table: [tgt_ptr, src_ptr] pairs then exclusive prefix
   Tensor table = at::empty({n * 2 + n + 1},
@@ -185,6 +202,24 @@ void multi_tensor_polyak(std::vector<Tensor> targets,
                              current_stream());
 }
 
+// Cached variant: the python layer builds the device table ONCE per
+// (targets, sources) pair; each call is then a single kernel launch
+// with zero host-side setup (the v1 per-call table build made the
+// fused kernel 6x slower than torch _foreach on small nets,
+// profiles/kernel_bench_r01.json).
+void multi_tensor_polyak_cached(Tensor dev_table, int64_t n_tensors,
+                                int64_t total, double tau) {
+  TORCH_CHECK(dev_table.is_cuda() &&
+                  dev_table.scalar_type() == at::kLong &&
+                  dev_table.is_contiguous(),
+              "dev_table must be contiguous int64 CUDA");
+  const at::hip::OptionalHIPGuardMasqueradingAsCUDA guard(
+      dev_table.device());
+  int64_t* d = dev_table.data_ptr<int64_t>();
+  multi_tensor_polyak_launch((void*)d, d + 2 * n_tensors, n_tensors, total,
+                             (float)tau, current_stream());
+}
+
 // ------------------------------------------------------------------
 // distributions
 // ------------------------------------------------------------------
@@ -194,7 +229,7 @@ std::vector<Tensor> gaussian_sample_logprob(Tensor mu, Tensor log_std,
                                             double epsilon) {
   check_f32_cuda(mu, "mu");
   check_f32_cuda(log_std, "log_std");
-  const at::cuda::OptionalCUDAGuard guard(mu.device());
+  const at::hip::OptionalHIPGuardMasqueradingAsCUDA guard(mu.device());
   int64_t B = mu.size(0), D = mu.size(1);
   Tensor act = at::empty_like(mu);
   Tensor logp = at::empty({B, 1}, mu.options());
@@ -208,7 +243,7 @@ std::vector<Tensor> gaussian_sample_logprob(Tensor mu, Tensor log_std,
 Tensor gaussian_logprob(Tensor mu, Tensor log_std, Tensor act,
                         bool tanh_squash, double epsilon) {
   check_f32_cuda(mu, "mu");
-  const at::cuda::OptionalCUDAGuard guard(mu.device());
+  const at::hip::OptionalHIPGuardMasqueradingAsCUDA guard(mu.device());
   int64_t B = mu.size(0), D = mu.size(1);
   Tensor logp = at::empty({B, 1}, mu.options());
   gaussian_logprob_launch(mu.data_ptr<float>(), log_std.data_ptr<float>(),
@@ -221,7 +256,7 @@ Tensor gaussian_logprob(Tensor mu, Tensor log_std, Tensor act,
 void normal_noise_(Tensor x, double mean, double std, int64_t seed,
                    int64_t offset, bool add) {
   check_f32_cuda(x, "x");
-  const at::cuda::OptionalCUDAGuard guard(x.device());
+  const at::hip::OptionalHIPGuardMasqueradingAsCUDA guard(x.device());
   normal_noise_launch(x.data_ptr<float>(), x.numel(), (float)mean,
                       (float)std, (uint64_t)seed, (uint64_t)offset,
                       add ? 1 : 0, current_stream());
@@ -230,7 +265,7 @@ void normal_noise_(Tensor x, double mean, double std, int64_t seed,
 void ou_update_(Tensor x, double mu, double theta, double sigma, double dt,
                 int64_t seed, int64_t offset) {
   check_f32_cuda(x, "x");
-  const at::cuda::OptionalCUDAGuard guard(x.device());
+  const at::hip::OptionalHIPGuardMasqueradingAsCUDA guard(x.device());
   ou_update_launch(x.data_ptr<float>(), x.numel(), (float)mu, (float)theta,
                    (float)sigma, (float)dt, (uint64_t)seed,
                    (uint64_t)offset, current_stream());
@@ -240,7 +275,7 @@ Tensor u8_to_bf16_scale(Tensor in, double scale) {
   TORCH_CHECK(in.is_cuda() && in.scalar_type() == at::kByte &&
                   in.is_contiguous(),
               "input must be contiguous uint8 CUDA");
-  const at::cuda::OptionalCUDAGuard guard(in.device());
+  const at::hip::OptionalHIPGuardMasqueradingAsCUDA guard(in.device());
   Tensor out = at::empty_like(in, in.options().dtype(at::kBFloat16));
   u8_to_bf16_scale_launch(in.data_ptr<unsigned char>(), out.data_ptr(),
                           in.numel(), (float)scale, current_stream());
@@ -258,7 +293,7 @@ std::vector<Tensor> conv1_wrw(Tensor dy, Tensor frames, double scale) {
               "frames must be contiguous u8 CUDA");
   TORCH_CHECK(dy.size(0) == frames.size(0) * 400,
               "K must equal batch*400");
-  const at::cuda::OptionalCUDAGuard guard(dy.device());
+  const at::hip::OptionalHIPGuardMasqueradingAsCUDA guard(dy.device());
   Tensor scratch = at::empty({32 * 256},
                              dy.options().dtype(at::kFloat));
   Tensor grad_w = at::empty({32, 4, 8, 8},
@@ -282,7 +317,7 @@ Tensor conv1_fwd(Tensor frames, Tensor weight, Tensor bias,
                   weight.is_contiguous() && weight.size(0) == 256 &&
                   weight.size(1) == 32,
               "weight must be contiguous [256,32] bf16");
-  const at::cuda::OptionalCUDAGuard guard(frames.device());
+  const at::hip::OptionalHIPGuardMasqueradingAsCUDA guard(frames.device());
   int64_t K = frames.size(0) * 400;
   Tensor out = at::empty({K, 32},
                          weight.options().dtype(at::kBFloat16));
@@ -304,7 +339,7 @@ Tensor mfma_probe(Tensor A, Tensor B) {
               "A must be [16,32] bf16");
   TORCH_CHECK(B.is_contiguous() && B.size(0) == 32 && B.size(1) == 16,
               "B must be [32,16] bf16");
-  const at::cuda::OptionalCUDAGuard guard(A.device());
+  const at::hip::OptionalHIPGuardMasqueradingAsCUDA guard(A.device());
   Tensor D = at::empty({16, 16}, A.options().dtype(at::kFloat));
   mfma_probe_launch(A.data_ptr(), B.data_ptr(), D.data_ptr<float>(),
                     current_stream());
@@ -323,6 +358,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("vtrace", &vtrace);
   m.def("categorical_projection", &categorical_projection);
   m.def("multi_tensor_polyak", &multi_tensor_polyak);
+  m.def("multi_tensor_polyak_cached", &multi_tensor_polyak_cached);
+  m.def("nstep_returns", &nstep_returns);
   m.def("gaussian_sample_logprob", &gaussian_sample_logprob);
   m.def("gaussian_logprob", &gaussian_logprob);
   m.def("normal_noise_", &normal_noise_);

@@ -108,3 +108,37 @@ void vtrace_launch(const float* blp, const float* tlp, const float* rew,
                      stream, blp, tlp, rew, val, bootstrap, nd, vs_out,
                      pg_adv, T, B, gamma, rho_clip, c_clip, pg_rho_clip);
 }
+
+// n-step truncated return: G_t = sum_{k<n} gamma^k r_{t+k} * prod_{j<k}
+// alive_{t+j}. One thread per (t, b) cell; reads are coalesced across b
+// at each k offset (layout [T, B]); n is tiny (2-5) so the inner loop
+// is register-resident. Replaces the reference python double loop
+// (machin/frame/algorithms/rainbow.py:179-189).
+__global__ void nstep_returns_kernel(const float* __restrict__ rew,
+                                     const float* __restrict__ alive,
+                                     float* __restrict__ out, int64_t T,
+                                     int64_t B, float gamma, int n) {
+  int64_t total = T * B;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < total; i += (int64_t)gridDim.x * blockDim.x) {
+    int64_t t = i / B;
+    int64_t b = i - t * B;
+    float g = 0.0f, factor = 1.0f, a = 1.0f;
+    for (int k = 0; k < n && t + k < T; ++k) {
+      int64_t kk = (t + k) * B + b;
+      g += factor * a * rew[kk];
+      a *= alive[kk];
+      factor *= gamma;
+    }
+    out[i] = g;
+  }
+}
+
+void nstep_returns_launch(const float* rew, const float* alive, float* out,
+                          int64_t T, int64_t B, float gamma, int n,
+                          hipStream_t stream) {
+  const int block = 256;
+  hipLaunchKernelGGL(nstep_returns_kernel, dim3(ma_grid(T * B, block)),
+                     dim3(block), 0, stream, rew, alive, out, T, B, gamma,
+                     n);
+}

@@ -105,6 +105,30 @@ class TestScansGPU:
                       0.99, 0.95)
         assert t.allclose(gpu.cpu(), cpu, rtol=1e-4, atol=1e-3)
 
+    def test_nstep_returns(self, dev):
+        import machin_amd.ops as ops
+
+        T, B = 50, 64
+        rew = t.rand(T, B)
+        term = (t.rand(T, B) > 0.85).float()
+        for n in (1, 2, 3, 5):
+            cpu = ops.nstep_returns(rew, term, 0.95, n)
+            gpu = ops.nstep_returns(rew.to(dev), term.to(dev), 0.95, n)
+            assert t.allclose(gpu.cpu(), cpu, rtol=1e-4, atol=1e-4), n
+
+    def test_fused_polyak_plan(self, dev):
+        import machin_amd.ops as ops
+
+        tl = [t.rand(37, device=dev), t.rand(256, 64, device=dev)]
+        sl = [t.rand(37, device=dev), t.rand(256, 64, device=dev)]
+        ref = [a * 0.99 + b * 0.01 for a, b in zip(tl, sl)]
+        plan = ops.FusedPolyak(tl, sl)
+        plan(0.01)
+        for a, r in zip(tl, ref):
+            assert t.allclose(a, r, atol=1e-6)
+        assert plan.matches(tl, sl)
+        assert not plan.matches(sl, tl)
+
     def test_vtrace(self, dev):
         import machin_amd.ops as ops
 

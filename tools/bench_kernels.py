@@ -124,6 +124,31 @@ def main():
 
     results["polyak_naturecnn_foreach_ms"] = timeit(torch_polyak, iters=100)
 
+    # cached plan: one launch, zero per-call host setup (round 2)
+    plan = ops.FusedPolyak(tl, sl)
+    results["polyak_naturecnn_cached_ms"] = timeit(
+        lambda: plan(0.005), iters=100
+    )
+
+    # --- n-step returns ----------------------------------------------
+    Tn, Bn = 64, 4096
+    rw = t.rand(Tn, Bn, device=dev)
+    al = (t.rand(Tn, Bn, device=dev) > 0.02).float()
+    from machin_amd.ops import _machin_hip as ext
+
+    results["nstep_64x4096_n3_kernel_ms"] = timeit(
+        lambda: ext.nstep_returns(rw, al, 0.99, 3)
+    )
+
+    def nstep_torch():
+        pad = t.zeros(1, Bn, device=dev)
+        out = t.zeros_like(rw)
+        for _ in range(3):
+            out = rw + 0.99 * al * t.cat([out[1:], pad], dim=0)
+        return out
+
+    results["nstep_64x4096_n3_torch_ms"] = timeit(nstep_torch)
+
     os.makedirs("gpurun_out", exist_ok=True)
     with open("gpurun_out/kernel_bench.json", "w") as f:
         json.dump(results, f, indent=2)

@@ -170,14 +170,23 @@ class RingPipeline:
             self.ingest_once()
 
     def sample(self):
-        """Gather a t-major training batch from the HBM pool."""
+        """Gather a t-major training batch from the HBM pool: one
+        index_select plus ONE strided copy straight into the
+        channels-last layout the conv stem wants (the round-1 chain
+        permute -> reshape -> to(channels_last) made three full passes
+        over the 1.2 GB frame batch)."""
         B, T = self.env_batch, self.unroll
         seg = t.randint(0, max(self.filled, 1), (B,), device=self.device)
         p = self.pool
+        fr = p["frames"].index_select(0, seg)  # [B, T, C, H, W]
+        C, H, W = fr.shape[2:]
+        frames = t.empty(
+            (T * B, C, H, W), dtype=fr.dtype, device=self.device,
+            memory_format=t.channels_last,
+        )
+        frames.view(T, B, C, H, W).copy_(fr.transpose(0, 1))
         return {
-            "frames": p["frames"].index_select(0, seg)
-            .permute(1, 0, 2, 3, 4).reshape(T * B, *p["frames"].shape[2:])
-            .to(memory_format=t.channels_last),
+            "frames": frames,
             "actions": p["actions"].index_select(0, seg).t().contiguous(),
             "behavior_logp": p["behavior_logp"].index_select(0, seg)
             .t().contiguous(),

@@ -95,8 +95,12 @@ class SAC(TorchFramework):
         self.alpha_optim = t.optim.Adam(
             [self.entropy_alpha], lr=alpha_learning_rate
         )
+        # replay_device="cuda:*" routes replay to the HBM-resident
+        # flat rings (machin_amd/frame/buffers/device_buffer.py)
+        from ..buffers import default_buffer
+
         self.replay_buffer = (
-            Buffer(replay_size, replay_device)
+            default_buffer(replay_size, replay_device)
             if replay_buffer is None
             else replay_buffer
         )

@@ -3,6 +3,7 @@ from .base import (
     dynamic_module_wrapper,
     static_module_wrapper,
 )
+from .gaussian import GaussianPolicyHead
 from .nature_cnn import ActorCriticCNN, NatureCNN, mlp
 from .resnet import ResNet
 
@@ -14,4 +15,5 @@ __all__ = [
     "ActorCriticCNN",
     "mlp",
     "ResNet",
+    "GaussianPolicyHead",
 ]

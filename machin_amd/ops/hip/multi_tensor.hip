@@ -16,7 +16,12 @@ struct MTEntry {
 };
 
 // chunk c covers elements [c*CHUNK, ...) of the virtual concatenation.
-#define MT_CHUNK 65536
+// CHUNK sets the grid: total/CHUNK blocks of MT_BLOCK threads. 4096
+// (16 elements/thread) puts a Nature-CNN-sized 1.7M-element list on
+// ~420 blocks — enough to spread over the 256 CUs; the round-1 value
+// of 65536 launched only 26 blocks and left the chip 90% idle
+// (0.102 ms vs 0.017 ms torch _foreach).
+#define MT_CHUNK 4096
 #define MT_BLOCK 256
 
 template <typename T>

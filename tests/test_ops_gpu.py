@@ -213,6 +213,74 @@ class TestDistributionsGPU:
         ).sum(dim=1, keepdim=True)
         assert t.allclose(logp, expect, atol=5e-3, rtol=1e-3)
 
+    def test_fused_head_gradients_match_torch(self, dev):
+        """The analytic backward of the fused tanh-Gaussian head must
+        equal autograd through the equivalent torch math (same
+        noise)."""
+        from machin_amd.model.nets.gaussian import _FusedTanhGaussian
+
+        t.manual_seed(3)
+        B, D = 64, 6
+        mu = t.randn(B, D, device=dev, requires_grad=True)
+        ls = (t.randn(B, D, device=dev) * 0.3).requires_grad_(True)
+        w1 = t.randn(B, D, device=dev)
+        w2 = t.randn(B, 1, device=dev)
+        a, logp = _FusedTanhGaussian.apply(mu, ls, 1234, 0, 1e-6)
+        ((a * w1).sum() + (logp * w2).sum()).backward()
+        g_mu, g_ls = mu.grad.clone(), ls.grad.clone()
+
+        # rebuild the same sample with torch ops (noise recovered
+        # from the kernel's output) and let autograd do the rest
+        with t.no_grad():
+            eps = (t.atanh(a.clamp(-1 + 1e-6, 1 - 1e-6)) - mu) \
+                * t.exp(-ls)
+        mu2 = mu.detach().clone().requires_grad_(True)
+        ls2 = ls.detach().clone().requires_grad_(True)
+        u2 = mu2 + t.exp(ls2) * eps
+        a2 = t.tanh(u2)
+        import math
+
+        logp2 = (
+            -0.5 * eps ** 2 - ls2 - 0.5 * math.log(2 * math.pi)
+            - t.log(1 - a2 * a2 + 1e-6)
+        ).sum(dim=1, keepdim=True)
+        assert t.allclose(a2, a, atol=1e-4)
+        assert t.allclose(logp2, logp, atol=1e-3)
+        ((a2 * w1).sum() + (logp2 * w2).sum()).backward()
+        assert t.allclose(mu2.grad, g_mu, rtol=1e-3, atol=1e-3)
+        assert t.allclose(ls2.grad, g_ls, rtol=1e-3, atol=1e-3)
+
+    def test_sac_trains_through_fused_head(self, dev):
+        """SAC end-to-end on GPU with the zoo GaussianActor: the
+        fused kernel is the sampling path (VERDICT next #5)."""
+        import torch.nn as nn
+
+        from machin_amd.auto.model_zoo import GaussianActor, QCritic
+        from machin_amd.frame.algorithms import SAC
+
+        frame = SAC(
+            GaussianActor(3, 1).to(dev),
+            QCritic(3, 1).to(dev), QCritic(3, 1).to(dev),
+            QCritic(3, 1).to(dev), QCritic(3, 1).to(dev),
+            t.optim.Adam, nn.MSELoss(),
+            replay_device=dev, batch_size=16,
+        )
+        for e in range(4):
+            ep = [
+                {
+                    "state": {"state": t.rand(1, 3)},
+                    "action": {"action": t.rand(1, 1) * 2 - 1},
+                    "next_state": {"state": t.rand(1, 3)},
+                    "reward": float(t.rand(1)),
+                    "terminal": i == 4,
+                }
+                for i in range(5)
+            ]
+            frame.store_episode(ep)
+        for _ in range(5):
+            out = frame.update()
+            assert all(v == v for v in out)
+
     def test_gaussian_logprob_of_given(self, dev):
         from machin_amd.ops import _require_ext
 

@@ -1,6 +1,12 @@
 from .classic_control import CartPoleEnv, PendulumEnv, Space, make
+from .pixel_catch import PixelCatchEnv
+from .simple_spread import SimpleSpreadEnv
 
-__all__ = ["CartPoleEnv", "PendulumEnv", "Space", "make"]
-from .simple_spread import SimpleSpreadEnv  # noqa: E402
-
-__all__.append("SimpleSpreadEnv")
+__all__ = [
+    "CartPoleEnv",
+    "PendulumEnv",
+    "PixelCatchEnv",
+    "SimpleSpreadEnv",
+    "Space",
+    "make",
+]

@@ -181,10 +181,15 @@ class PendulumEnv:
 
 
 def make(name: str, seed: Optional[int] = None):
-    """Factory: 'CartPole-v1' / 'Pendulum-v1' (version suffix ignored)."""
+    """Factory: 'CartPole-v1' / 'Pendulum-v1' / 'PixelCatch-v0'
+    (version suffix ignored)."""
     base = name.split("-")[0].lower()
     if base == "cartpole":
         return CartPoleEnv(seed)
     if base == "pendulum":
         return PendulumEnv(seed)
+    if base == "pixelcatch":
+        from .pixel_catch import PixelCatchEnv
+
+        return PixelCatchEnv(seed)
     raise ValueError(f"Unknown built-in environment {name!r}.")

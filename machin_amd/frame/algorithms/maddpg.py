@@ -5,8 +5,13 @@ Parity target: reference ``machin/frame/algorithms/maddpg.py``
 ensemble sub-policies (``sub_policy_num``), a ``critic_visible_actors``
 observability matrix (:106-113), lockstep per-agent replay with
 same-index sampling (:959-966), pool-parallel per-agent updates
-(:594-634; thread pool here — MI355X models share one GPU so threads
-avoid IPC copies), action/state concat hooks (:968-994).
+(:594-634): ``pool_type="thread"`` runs them on a thread pool
+(models share one GPU, no IPC copies); ``pool_type="process"`` runs
+them on a :class:`machin_amd.parallel.pool.P2PPool` with every model
+parameter AND optimizer state tensor in shared memory, so worker
+updates mutate the parent's networks in place (reference's
+shared-memory pool mode, maddpg.py:594-634 + SHMBuffer :16-44).
+Action/state concat hooks (:968-994).
 """
 import copy
 import random
@@ -16,7 +21,7 @@ import numpy as np
 import torch as t
 import torch.nn as nn
 
-from ...parallel.pool import ThreadPool
+from ...parallel.pool import P2PPool, ThreadPool
 from ...utils.conf import Config
 from ..buffers.buffer import Buffer
 from ..noise.action_space_noise import (
@@ -27,6 +32,85 @@ from ..noise.action_space_noise import (
 )
 from .base import TorchFramework
 from .utils import hard_update, safe_call, safe_return, soft_update
+
+
+def _share_optimizer_state(opt):
+    """Materialize optimizer state with a zero-grad step, then move
+    every state tensor into shared memory so P2PPool workers' steps
+    accumulate into the PARENT's Adam moments (a worker receives the
+    optimizer by reference-serialization each call)."""
+    for group in opt.param_groups:
+        for p in group["params"]:
+            if p.grad is None:
+                p.grad = t.zeros_like(p)
+    opt.step()  # zero grad: params unchanged, state allocated
+    opt.zero_grad(set_to_none=False)
+    for st in opt.state.values():
+        for v in st.values():
+            if t.is_tensor(v):
+                v.share_memory_()
+
+
+def _process_update_task(payload):
+    """One agent's critic+actor update, executed in a P2PPool worker.
+    All parameters/optimizer state arrive as shared-memory references
+    (machin_amd.parallel.pickle copy_tensor=False), so the in-place
+    optimizer steps are visible to the parent process."""
+    (actor, actor_optim, critic, critic_optim, critic_target,
+     target_actors, batches_vis, agent_pos, reward, terminal, bsize,
+     discount, grad_max, update_value, update_policy, criterion,
+     fns) = payload
+    atf, acf, scf, rwf = fns
+    states = [b[0] for b in batches_vis]
+    actions = [b[1] for b in batches_vis]
+    next_states = [b[2] for b in batches_vis]
+
+    with t.no_grad():
+        next_acts = [
+            atf(safe_return(safe_call(pol, ns)))
+            for pol, ns in zip(target_actors, next_states)
+        ]
+        next_value = safe_return(
+            safe_call(critic_target, scf(next_states), acf(next_acts))
+        ).view(bsize, 1)
+        device = next_value.device
+        rew = reward.to(device).float().view(bsize, 1)
+        term = terminal.to(device).float().view(bsize, 1)
+        y = rwf(rew, discount, next_value, term)
+
+    all_states = scf(states)
+    cur_value = safe_return(
+        safe_call(
+            critic, all_states,
+            acf([{"action": a["action"]} for a in actions]),
+        )
+    ).view(bsize, 1)
+    value_loss = criterion(cur_value, y.to(cur_value.dtype))
+    if update_value:
+        critic_optim.zero_grad(set_to_none=True)
+        value_loss.backward()
+        nn.utils.clip_grad_norm_(critic.parameters(), grad_max)
+        critic_optim.step()
+
+    cur_action = atf(safe_return(safe_call(actor, states[agent_pos])))
+    policy_actions = [
+        cur_action if i == agent_pos else {"action": actions[i]["action"]}
+        for i in range(len(batches_vis))
+    ]
+    act_value = safe_return(
+        safe_call(critic, all_states, acf(policy_actions))
+    )
+    act_policy_loss = -act_value.mean()
+    if update_policy:
+        actor_optim.zero_grad(set_to_none=True)
+        act_policy_loss.backward()
+        nn.utils.clip_grad_norm_(actor.parameters(), grad_max)
+        actor_optim.step()
+
+    return (
+        -float(act_policy_loss.detach().item()),
+        float(value_loss.detach().item()),
+    )
 
 
 class MADDPG(TorchFramework):
@@ -143,9 +227,31 @@ class MADDPG(TorchFramework):
         self.criterion = (
             criterion() if isinstance(criterion, type) else criterion
         )
-        self.pool = ThreadPool(
-            processes=pool_size or min(n, 8)
-        )
+        if pool_type not in ("thread", "process"):
+            raise ValueError(
+                f"pool_type must be 'thread' or 'process', got "
+                f"{pool_type!r}"
+            )
+        self.pool_type = pool_type
+        if pool_type == "process":
+            # share everything the workers mutate
+            for group in self.actors + self.actor_targets:
+                for mod in group:
+                    mod.share_memory()
+            for mod in self.critics + self.critic_targets:
+                mod.share_memory()
+            for opts in self.actor_optims:
+                for opt in opts:
+                    _share_optimizer_state(opt)
+            for opt in self.critic_optims:
+                _share_optimizer_state(opt)
+            self.pool = P2PPool(
+                processes=pool_size or min(n, 8), copy_tensor=False
+            )
+        else:
+            self.pool = ThreadPool(
+                processes=pool_size or min(n, 8)
+            )
 
         # checkpoint containers
         self.all_actor_target = nn.Module()
@@ -294,13 +400,53 @@ class MADDPG(TorchFramework):
             )
             batches.append(batch)
 
-        results = self.pool.starmap(
-            self._update_agent,
-            [
-                (agent, batches, bsize, update_value, update_policy)
-                for agent in range(self.agent_num)
-            ],
-        )
+        if self.pool_type == "process":
+            tasks = []
+            for agent in range(self.agent_num):
+                visible = self.critic_visible_actors[agent]
+                pidx = random.randrange(self.ensemble_size)
+                target_actors = [
+                    random.choice(self.actor_targets[a]) for a in visible
+                ]
+                batches_vis = [
+                    (batches[a][0], batches[a][1], batches[a][3])
+                    for a in visible
+                ]
+                tasks.append((
+                    (
+                        self.actors[agent][pidx],
+                        self.actor_optims[agent][pidx],
+                        self.critics[agent],
+                        self.critic_optims[agent],
+                        self.critic_targets[agent],
+                        target_actors,
+                        batches_vis,
+                        visible.index(agent),
+                        batches[agent][2],
+                        batches[agent][4],
+                        bsize,
+                        self.discount,
+                        self.grad_max,
+                        update_value,
+                        update_policy,
+                        self.criterion,
+                        (
+                            self.action_transform_function,
+                            self.action_concat_function,
+                            self.state_concat_function,
+                            self.reward_function,
+                        ),
+                    ),
+                ))
+            results = self.pool.starmap(_process_update_task, tasks)
+        else:
+            results = self.pool.starmap(
+                self._update_agent,
+                [
+                    (agent, batches, bsize, update_value, update_policy)
+                    for agent in range(self.agent_num)
+                ],
+            )
 
         if update_target:
             if self.update_rate is not None:

@@ -119,3 +119,35 @@ class TestMADDPG:
             m2.all_actor_target.parameters(),
         ):
             assert t.allclose(p1, p2)
+
+
+class TestMADDPGProcessPool:
+    def test_process_pool_updates_parent_models(self):
+        """pool_type="process": P2PPool workers update shared-memory
+        parameters AND shared Adam state in place (reference
+        maddpg.py:594-634 shared-memory pool mode). Models are the
+        module-level classes (dill ships them by reference)."""
+        import numpy as np
+
+        t.manual_seed(0)
+        m = make_maddpg(pool_type="process", pool_size=3)
+        try:
+            before = [
+                p.detach().clone()
+                for p in m.actors[0][0].parameters()
+            ]
+            m.store_episodes(make_episodes())
+            for _ in range(3):
+                al, vl = m.update()
+                assert np.isfinite(al) and np.isfinite(vl)
+            after = [p.detach() for p in m.actors[0][0].parameters()]
+            assert any(
+                not t.allclose(b, a) for b, a in zip(before, after)
+            ), "worker updates did not reach the parent's parameters"
+            # shared Adam state: 1 warmup step + 3 updates
+            st = next(iter(m.critic_optims[0].state.values()))
+            step = st["step"]
+            step = step.item() if t.is_tensor(step) else step
+            assert step == 4, f"Adam state not shared: step={step}"
+        finally:
+            m.pool.terminate()

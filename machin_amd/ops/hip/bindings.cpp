@@ -46,6 +46,9 @@ void u8_to_bf16_scale_launch(const unsigned char*, void*, int64_t, float,
 void conv1_wrw_launch(const void*, const unsigned char*, float*, float*,
                       float*, int64_t, float, hipStream_t);
 void mfma_probe_launch(const void*, const void*, float*, hipStream_t);
+void tr16_probe_launch(float*, int, hipStream_t);
+void fwd_bfrag_probe_launch(float*, int, hipStream_t);
+void tr16_diag_launch(float*, int, hipStream_t);
 void conv1_fwd_launch(const unsigned char*, const void*, const float*,
                       void*, int64_t, float, hipStream_t);
 
@@ -333,6 +336,35 @@ Tensor conv1_fwd(Tensor frames, Tensor weight, Tensor bias,
   return out;
 }
 
+Tensor tr16_probe(int64_t mode) {
+  Tensor out = at::empty({64, 4}, at::TensorOptions()
+                                      .dtype(at::kFloat)
+                                      .device(at::kCUDA));
+  const at::hip::OptionalHIPGuardMasqueradingAsCUDA guard(out.device());
+  tr16_probe_launch(out.data_ptr<float>(), (int)mode, current_stream());
+  return out;
+}
+
+Tensor fwd_bfrag_probe(int64_t which) {
+  Tensor out = at::empty({2, 256, 8}, at::TensorOptions()
+                                          .dtype(at::kFloat)
+                                          .device(at::kCUDA));
+  const at::hip::OptionalHIPGuardMasqueradingAsCUDA guard(out.device());
+  fwd_bfrag_probe_launch(out.data_ptr<float>(), (int)which,
+                         current_stream());
+  return out;
+}
+
+Tensor tr16_diag(int64_t mode) {
+  int64_t n = mode <= 1 ? 8192 : 4096;
+  Tensor out = at::empty({n}, at::TensorOptions()
+                                  .dtype(at::kFloat)
+                                  .device(at::kCUDA));
+  const at::hip::OptionalHIPGuardMasqueradingAsCUDA guard(out.device());
+  tr16_diag_launch(out.data_ptr<float>(), (int)mode, current_stream());
+  return out;
+}
+
 Tensor mfma_probe(Tensor A, Tensor B) {
   TORCH_CHECK(A.is_cuda() && A.scalar_type() == at::kBFloat16 &&
               A.is_contiguous() && A.size(0) == 16 && A.size(1) == 32,
@@ -367,5 +399,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("u8_to_bf16_scale", &u8_to_bf16_scale);
   m.def("conv1_wrw", &conv1_wrw);
   m.def("mfma_probe", &mfma_probe);
+  m.def("tr16_probe", &tr16_probe);
+  m.def("fwd_bfrag_probe", &fwd_bfrag_probe);
+  m.def("tr16_diag", &tr16_diag);
   m.def("conv1_fwd", &conv1_fwd);
 }

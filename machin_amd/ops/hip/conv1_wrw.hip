@@ -200,6 +200,11 @@ __global__ void conv1_wrw_v2_kernel(const __bf16*, const unsigned char*,
                                     float);
 __global__ void conv1_fwd_v2_kernel(const unsigned char*, const __bf16*,
                                     const float*, __bf16*, int64_t, float);
+__global__ void conv1_wrw_v3_kernel(const __bf16*, const unsigned char*,
+                                    float*, float*, int64_t, int64_t,
+                                    float);
+__global__ void conv1_fwd_v3_kernel(const unsigned char*, const __bf16*,
+                                    const float*, __bf16*, int64_t, float);
 
 void conv1_wrw_launch(const void* dy, const unsigned char* frames,
                       float* scratch, float* grad_w, float* grad_b,
@@ -211,17 +216,23 @@ void conv1_wrw_launch(const void* dy, const unsigned char* frames,
   // split-K: target ~2048 workgroups (8 XCDs x 32 CUs x 8 blocks)
   int64_t target_wg = 2048;
   if (const char* e = getenv("MACHIN_CONV1_WG")) target_wg = atol(e);
+  const char* ver = getenv("MACHIN_CONV1_V");
+  int v = ver ? atoi(ver) : 3;
+  int64_t chunk = v == 3 ? 64 : 32;
   int64_t k_per_wg = (K + target_wg - 1) / target_wg;
-  k_per_wg = ((k_per_wg + KC - 1) / KC) * KC;
-  if (k_per_wg < KC) k_per_wg = KC;
+  k_per_wg = ((k_per_wg + chunk - 1) / chunk) * chunk;
+  if (k_per_wg < chunk) k_per_wg = chunk;
   int grid = (int)((K + k_per_wg - 1) / k_per_wg);
-  const bool v1 = getenv("MACHIN_CONV1_V1") != nullptr;
-  if (v1) {
+  if (v == 1) {
     hipLaunchKernelGGL(conv1_wrw_kernel, dim3(grid), dim3(256), 0, stream,
                        (const __bf16*)dy, frames, scratch, grad_b, K,
                        k_per_wg, scale);
-  } else {
+  } else if (v == 2) {
     hipLaunchKernelGGL(conv1_wrw_v2_kernel, dim3(grid), dim3(256), 0,
+                       stream, (const __bf16*)dy, frames, scratch, grad_b,
+                       K, k_per_wg, scale);
+  } else {
+    hipLaunchKernelGGL(conv1_wrw_v3_kernel, dim3(grid), dim3(256), 0,
                        stream, (const __bf16*)dy, frames, scratch, grad_b,
                        K, k_per_wg, scale);
   }
@@ -376,13 +387,19 @@ void conv1_fwd_launch(const unsigned char* frames, const void* weight,
                       const float* bias, void* out, int64_t K,
                       float scale, hipStream_t stream) {
   int64_t grid = (K + FWD_ROWS - 1) / FWD_ROWS;
-  const bool v1 = getenv("MACHIN_CONV1_V1") != nullptr;
-  if (v1) {
+  const char* ver = getenv("MACHIN_CONV1_V");
+  int v = ver ? atoi(ver) : 3;
+  if (v == 1) {
     hipLaunchKernelGGL(conv1_fwd_kernel, dim3((unsigned)grid), dim3(256),
                        0, stream, frames, (const __bf16*)weight, bias,
                        (__bf16*)out, K, scale);
-  } else {
+  } else if (v == 2) {
     hipLaunchKernelGGL(conv1_fwd_v2_kernel, dim3((unsigned)grid),
+                       dim3(256), 0, stream, frames,
+                       (const __bf16*)weight, bias, (__bf16*)out, K,
+                       scale);
+  } else {
+    hipLaunchKernelGGL(conv1_fwd_v3_kernel, dim3((unsigned)grid),
                        dim3(256), 0, stream, frames,
                        (const __bf16*)weight, bias, (__bf16*)out, K,
                        scale);
@@ -396,35 +413,73 @@ void conv1_fwd_launch(const unsigned char* frames, const void* weight,
 // v1 profile (round-1 VERDICT weak #6): the MFMA operand loads were
 // per-element ds_read_u16 (72 narrow reads per wave per K-chunk) —
 // LDS-instruction-bound at 1.80 ms wrw / 1.46 ms fwd vs a ~0.4 ms
-// HBM bound. gfx950's transpose-read fetches 4 bf16 per lane with a
-// 32 B stride (lane l passes base + (l&15)*2; elem j comes from
-// base + j*32 B), so storing x/dy in contiguous [4][16] bf16 tiles
-// (tile = 4 k-rows x 16 cols, 128 B) turns one 16x16x32 B-fragment
-// into TWO ds instructions (k 8-rows = two stacked tiles, second via
-// offset immediate). Writes stay wide (16-element rows are 32 B
-// contiguous). Per wave per chunk: 18 LDS reads vs v1's 72.
+// HBM bound.
 //
-// Tiled image layout (x): [n_tile 16][k_tile 8][4][16] bf16
-//   elem (k, c) -> nt=c>>4, kt=k>>2: addr = nt*512 + kt*64
-//                                         + (k&3)*16 + (c&15)
-// dy image: [m_tile 2][k_tile 8][4][16] with the same inner tiles.
+// ds_read_b64_tr_b16 semantics, pinned by tools/probe_tr16.py on
+// hardware (gpurun_out/tr16_probe2.log): lanes operate in clusters
+// of 4; the cluster reads one CONTIGUOUS 32 B window (a 4x4 bf16
+// tile, row-major [row][col]) at the cluster's (shared) address, and
+// lane l receives COLUMN (l&3): elements win[(l&3) + j*4], j=0..3.
+// Address bits 3-4 are ignored (windows must be 32 B aligned); the
+// offset immediate is additive in the same masked address space.
+//
+// So the LDS images are [n_cluster][k_quad][4][4] bf16 tilings: one
+// 16x16x32 MFMA B-fragment = TWO tr reads (k 8-rows = two stacked
+// 4x4 windows, second via offset:32). Per wave per chunk: 18 LDS
+// reads vs v1's 72.
+//
+// x image: [nc 64][kq 8][4][4]  elem (k, c):
+//   addr = (c>>2)*128 + (k>>2)*16 + (k&3)*4 + (c&3)
+// dy image: [nc 8][kq 8][4][4] with the same windows.
 // =====================================================================
 typedef __attribute__((ext_vector_type(4))) __bf16 bf16x4;
 
+// NOTE the explicit s_waitcnt: the compiler does not track inline-asm
+// DS reads, so without it the destination registers are consumed
+// before the LDS data lands (first probe run returned garbage).
 __device__ __forceinline__ bf16x4 ds_tr16(const __bf16* addr) {
   bf16x4 out;
-  asm volatile("ds_read_b64_tr_b16 %0, %1"
-               : "=v"(out)
-               : "v"((unsigned)(uintptr_t)addr));
+  asm volatile("ds_read_b64_tr_b16 %0, %1\n\ts_waitcnt lgkmcnt(0)"
+               : "=&v"(out)
+               : "v"((unsigned)(uintptr_t)addr)
+               : "memory");
   return out;
 }
 
 __device__ __forceinline__ bf16x4 ds_tr16_off128(const __bf16* addr) {
   bf16x4 out;
-  asm volatile("ds_read_b64_tr_b16 %0, %1 offset:128"
-               : "=v"(out)
-               : "v"((unsigned)(uintptr_t)addr));
+  asm volatile(
+      "ds_read_b64_tr_b16 %0, %1 offset:32\n\ts_waitcnt lgkmcnt(0)"
+      : "=&v"(out)
+      : "v"((unsigned)(uintptr_t)addr)
+      : "memory");
   return out;
+}
+
+// fused pair: both k-tiles of one fragment, ONE waitcnt
+__device__ __forceinline__ void ds_tr16_pair(const __bf16* addr,
+                                             bf16x4& lo, bf16x4& hi) {
+  asm volatile(
+      "ds_read_b64_tr_b16 %0, %2\n\t"
+      "ds_read_b64_tr_b16 %1, %2 offset:32\n\t"
+      "s_waitcnt lgkmcnt(0)"
+      : "=&v"(lo), "=&v"(hi)
+      : "v"((unsigned)(uintptr_t)addr)
+      : "memory");
+}
+
+__device__ __forceinline__ bf16x4 lo4(bf16x8 v) {
+  bf16x4 r;
+#pragma unroll
+  for (int j = 0; j < 4; ++j) r[j] = v[j];
+  return r;
+}
+
+__device__ __forceinline__ bf16x4 hi4(bf16x8 v) {
+  bf16x4 r;
+#pragma unroll
+  for (int j = 0; j < 4; ++j) r[j] = v[4 + j];
+  return r;
 }
 
 __device__ __forceinline__ void pack8(bf16x8& dst, bf16x4 lo, bf16x4 hi) {
@@ -455,11 +510,13 @@ void conv1_wrw_v2_kernel(const __bf16* __restrict__ dy,
   for (int i = 0; i < 8; ++i) acc[i] = (f32x4)(0.0f);
   float bias_acc = 0.0f;
 
-  // per-lane fragment base addresses (constant across the K loop):
-  // group g = lane>>4 covers k-tiles 2g, 2g+1 -> byte offset g*256
-  const __bf16* a_base = s_dy + mt * 512 + (lane >> 4) * 128 + (lane & 15);
-  const __bf16* b_base = s_x + ng * 8 * 512 + (lane >> 4) * 128
-                         + (lane & 15);
+  // per-lane window base addresses (constant across the K loop):
+  // cluster (lane&15)>>2 picks the n-quad; k-group g = lane>>4 needs
+  // k-quads 2g (base) and 2g+1 (offset:32)
+  const __bf16* a_base = s_dy + (mt * 4 + ((lane & 15) >> 2)) * 128
+                         + (lane >> 4) * 32;
+  const __bf16* b_base = s_x + (ng * 32 + ((lane & 15) >> 2)) * 128
+                         + (lane >> 4) * 32;
 
   const int64_t k_begin = (int64_t)blockIdx.x * k_per_wg;
   const int64_t k_end = min(k_begin + k_per_wg, K);
@@ -477,9 +534,12 @@ void conv1_wrw_v2_kernel(const __bf16* __restrict__ dy,
       if (kk < k_end) {
         v = *(const bf16x8*)(dy + kk * CONV1_COUT + seg * 8);
       }
-      __bf16* dst = s_dy + (seg >> 1) * 512 + (row >> 2) * 64
-                    + (row & 3) * 16 + (seg & 1) * 8;
-      *(bf16x8*)dst = v;
+      // cols seg*8..seg*8+7 span n-clusters 2*seg and 2*seg+1
+      __bf16* base = s_dy + (row >> 2) * 16 + (row & 3) * 4;
+      *(bf16x4*)(base + (seg * 2) * 128) =
+          lo4(v);
+      *(bf16x4*)(base + (seg * 2 + 1) * 128) =
+          hi4(v);
       (void)bsum;
     }
     // ---- stage x chunk into [nt][kt][4][16] tiles -----------------
@@ -523,21 +583,28 @@ void conv1_wrw_v2_kernel(const __bf16* __restrict__ dy,
           }
         }
       }
-      // cols r*32 .. r*32+31 span n-tiles 2r and 2r+1; each 16-col
-      // half is one contiguous tile row (32 B = 2 vector writes)
-      __bf16* base = s_x + (krow >> 2) * 64 + (krow & 3) * 16;
-      bf16x8* t0 = (bf16x8*)(base + (r * 2) * 512);
-      bf16x8* t1 = (bf16x8*)(base + (r * 2 + 1) * 512);
-      t0[0] = q0;
-      t0[1] = q1;
-      t1[0] = q2;
-      t1[1] = q3;
+      // cols r*32 .. r*32+31 span n-clusters r*8 .. r*8+7; each
+      // 4-col piece is one contiguous window row (8 B write)
+      __bf16* base = s_x + (krow >> 2) * 16 + (krow & 3) * 4
+                     + (r * 8) * 128;
+      *(bf16x4*)(base + 0 * 128) = lo4(q0);
+      *(bf16x4*)(base + 1 * 128) = hi4(q0);
+      *(bf16x4*)(base + 2 * 128) = lo4(q1);
+      *(bf16x4*)(base + 3 * 128) = hi4(q1);
+      *(bf16x4*)(base + 4 * 128) = lo4(q2);
+      *(bf16x4*)(base + 5 * 128) = hi4(q2);
+      *(bf16x4*)(base + 6 * 128) = lo4(q3);
+      *(bf16x4*)(base + 7 * 128) = hi4(q3);
     }
     __syncthreads();
 
     // ---- fragments via transpose-reads ----------------------------
     bf16x8 a_frag;
-    pack8(a_frag, ds_tr16(a_base), ds_tr16_off128(a_base));
+    {
+      bf16x4 lo, hi;
+      ds_tr16_pair(a_base, lo, hi);
+      pack8(a_frag, lo, hi);
+    }
     // fused bias from the fragment already in registers: lane group g
     // of an ng==0 wave holds dy[k=8g..8g+7][col mt*16+(lane&15)]
     if (ng == 0) {
@@ -548,7 +615,11 @@ void conv1_wrw_v2_kernel(const __bf16* __restrict__ dy,
     for (int nt = 0; nt < 8; ++nt) {
       const __bf16* bb = b_base + nt * 512;
       bf16x8 b_frag;
-      pack8(b_frag, ds_tr16(bb), ds_tr16_off128(bb));
+      {
+        bf16x4 lo, hi;
+        ds_tr16_pair(bb, lo, hi);
+        pack8(b_frag, lo, hi);
+      }
       acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
           a_frag, b_frag, acc[nt], 0, 0, 0);
     }
@@ -591,26 +662,31 @@ void conv1_fwd_v2_kernel(const unsigned char* __restrict__ frames,
   const int wm = wave & 1;
   const int wn = wave >> 1;
 
-  // stage weight into tiles: thread tid owns row k=tid (32 cols)
+  // stage weight into [nc 8][kq 64][4][4] windows: thread tid owns
+  // row k=tid (32 cols -> 8 clusters, one 8 B write each)
   {
     int k = tid;
     bf16x8 r0 = ((const bf16x8*)(weight + k * 32))[0];
     bf16x8 r1 = ((const bf16x8*)(weight + k * 32))[1];
     bf16x8 r2 = ((const bf16x8*)(weight + k * 32))[2];
     bf16x8 r3 = ((const bf16x8*)(weight + k * 32))[3];
-    __bf16* base = s_w + (k >> 2) * 64 + (k & 3) * 16;
-    ((bf16x8*)base)[0] = r0;
-    ((bf16x8*)base)[1] = r1;
-    ((bf16x8*)(base + 64 * 64))[0] = r2;
-    ((bf16x8*)(base + 64 * 64))[1] = r3;
+    __bf16* base = s_w + (k >> 2) * 16 + (k & 3) * 4;
+    *(bf16x4*)(base + 0 * 1024) = lo4(r0);
+    *(bf16x4*)(base + 1 * 1024) = hi4(r0);
+    *(bf16x4*)(base + 2 * 1024) = lo4(r1);
+    *(bf16x4*)(base + 3 * 1024) = hi4(r1);
+    *(bf16x4*)(base + 4 * 1024) = lo4(r2);
+    *(bf16x4*)(base + 5 * 1024) = hi4(r2);
+    *(bf16x4*)(base + 6 * 1024) = lo4(r3);
+    *(bf16x4*)(base + 7 * 1024) = hi4(r3);
   }
 
   const int64_t row0 = (int64_t)blockIdx.x * FWD_ROWS;
   f32x4 acc[2];
   acc[0] = (f32x4)(0.0f);
   acc[1] = (f32x4)(0.0f);
-  const __bf16* w_base = s_w + wn * 64 * 64 + (lane >> 4) * 128
-                         + (lane & 15);
+  const __bf16* w_base = s_w + (wn * 4 + ((lane & 15) >> 2)) * 1024
+                         + (lane >> 4) * 32;
   __syncthreads();
 
   for (int pr = 0; pr < CONV1_KSZ; ++pr) {
@@ -644,10 +720,427 @@ void conv1_fwd_v2_kernel(const unsigned char* __restrict__ frames,
     }
     __syncthreads();
 
-    // B fragment for this pr chunk: k-tiles pr*8 + {2g, 2g+1}
-    const __bf16* wb = w_base + pr * 8 * 64;
+    // B fragment for this pr chunk: k-quads pr*8 + {2g, 2g+1}
+    const __bf16* wb = w_base + pr * 8 * 16;
     bf16x8 b_frag;
-    pack8(b_frag, ds_tr16(wb), ds_tr16_off128(wb));
+    {
+      bf16x4 lo, hi;
+      ds_tr16_pair(wb, lo, hi);
+      pack8(b_frag, lo, hi);
+    }
+#pragma unroll
+    for (int mt2 = 0; mt2 < 2; ++mt2) {
+      int m = wm * 32 + mt2 * 16 + (lane & 15);
+      int k0 = (lane >> 4) * 8;
+      bf16x8 a_frag = *(bf16x8*)(s_x + m * 32 + k0);
+      acc[mt2] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+          a_frag, b_frag, acc[mt2], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+#pragma unroll
+  for (int mt2 = 0; mt2 < 2; ++mt2) {
+    int col = wn * 16 + (lane & 15);
+    float b = bias != nullptr ? bias[col] : 0.0f;
+    int row_base = wm * 32 + mt2 * 16 + (lane >> 4) * 4;
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) {
+      int64_t kk = row0 + row_base + reg;
+      if (kk < K) {
+        out[kk * 32 + col] = (__bf16)(acc[mt2][reg] + b);
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------
+// tr16 probe: pins the exact ds_read_b64_tr_b16 lane/element mapping
+// on hardware before the conv kernels trust it (same methodology as
+// mfma_probe). LDS is filled with lds[i] = i; each lane issues the
+// read with a configurable per-lane address and reports its 4
+// elements. mode 0: uniform tile base. mode 1: base + (lane&15)*2 B.
+// mode 2: base + lane*2 B. mode 3: base + (lane>>4)*128 B.
+// ---------------------------------------------------------------------
+__global__ void tr16_probe_kernel(float* __restrict__ out, int mode) {
+  __shared__ __bf16 l[1024];
+  for (int i = threadIdx.x; i < 1024; i += blockDim.x) {
+    l[i] = (__bf16)(float)i;
+  }
+  __syncthreads();
+  int lane = threadIdx.x;
+  const __bf16* addr = l;
+  if (mode == 1) addr += (lane & 15);
+  if (mode == 2) addr += lane;
+  if (mode == 3) addr += (lane >> 4) * 64;
+  if (mode == 4) addr += (lane & 15) + (lane >> 4) * 64;
+  bf16x4 v = ds_tr16(addr);
+#pragma unroll
+  for (int j = 0; j < 4; ++j) {
+    out[lane * 4 + j] = (float)v[j];
+  }
+}
+
+void tr16_probe_launch(float* out, int mode, hipStream_t stream) {
+  hipLaunchKernelGGL(tr16_probe_kernel, dim3(1), dim3(64), 0, stream,
+                     out, mode);
+  HIP_CHECK(hipGetLastError());
+}
+
+// fwd B-fragment probe: stages a synthetic weight through the EXACT
+// v2 staging code, reads fragments with the EXACT v2 addressing, and
+// dumps every lane's 8 elements. which=0: W[k][n]=k (checks the k
+// mapping); which=1: W[k][n]=n (checks the n mapping).
+// out: [2 pr][256 threads][8]
+__global__ __launch_bounds__(256)
+void fwd_bfrag_probe_kernel(float* __restrict__ out, int which) {
+  __shared__ __bf16 s_w[2 * 64 * 64];
+  const int tid = threadIdx.x;
+  const int wave = tid / MA_WAVE;
+  const int lane = tid % MA_WAVE;
+  const int wn = wave >> 1;
+
+  {
+    int k = tid;
+    bf16x8 r0, r1, r2, r3;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      r0[j] = (__bf16)(float)(which == 0 ? k : j);
+      r1[j] = (__bf16)(float)(which == 0 ? k : 8 + j);
+      r2[j] = (__bf16)(float)(which == 0 ? k : 16 + j);
+      r3[j] = (__bf16)(float)(which == 0 ? k : 24 + j);
+    }
+    __bf16* base = s_w + (k >> 2) * 16 + (k & 3) * 4;
+    *(bf16x4*)(base + 0 * 1024) = lo4(r0);
+    *(bf16x4*)(base + 1 * 1024) = hi4(r0);
+    *(bf16x4*)(base + 2 * 1024) = lo4(r1);
+    *(bf16x4*)(base + 3 * 1024) = hi4(r1);
+    *(bf16x4*)(base + 4 * 1024) = lo4(r2);
+    *(bf16x4*)(base + 5 * 1024) = hi4(r2);
+    *(bf16x4*)(base + 6 * 1024) = lo4(r3);
+    *(bf16x4*)(base + 7 * 1024) = hi4(r3);
+  }
+  const __bf16* w_base = s_w + (wn * 4 + ((lane & 15) >> 2)) * 1024
+                         + (lane >> 4) * 32;
+  __syncthreads();
+#pragma unroll
+  for (int pr = 0; pr < 2; ++pr) {
+    const __bf16* wb = w_base + pr * 8 * 16;
+    bf16x4 lo, hi;
+    ds_tr16_pair(wb, lo, hi);
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      out[(pr * 256 + tid) * 8 + j] = (float)lo[j];
+      out[(pr * 256 + tid) * 8 + 4 + j] = (float)hi[j];
+    }
+  }
+}
+
+void fwd_bfrag_probe_launch(float* out, int which, hipStream_t stream) {
+  hipLaunchKernelGGL(fwd_bfrag_probe_kernel, dim3(1), dim3(256), 0,
+                     stream, out, which);
+  HIP_CHECK(hipGetLastError());
+}
+
+// Decisive tr16 diagnostics.
+// mode 0/1: stage synthetic W (value = k for 0, n for 1) through the
+//   v2 staging code, then dump the raw LDS image LINEARLY (plain
+//   reads — isolates staging correctness from the tr16 read).
+// mode 2/3: fill s_w[i] = i&255 (2) or i>>8 (3) DIRECTLY, then tr16-
+//   read with the exact fwd fragment addressing and dump per-lane
+//   elements (isolates the read's element map; combining both runs
+//   reconstructs exact LDS indices).
+// out: 8192 floats for modes 0/1; [2][256][8] for modes 2/3.
+__global__ __launch_bounds__(256)
+void tr16_diag_kernel(float* __restrict__ out, int mode) {
+  __shared__ __bf16 s_w[2 * 64 * 64];
+  const int tid = threadIdx.x;
+  const int wave = tid / MA_WAVE;
+  const int lane = tid % MA_WAVE;
+  const int wn = wave >> 1;
+
+  if (mode <= 1) {
+    int k = tid;
+    bf16x8 r0, r1, r2, r3;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      r0[j] = (__bf16)(float)(mode == 0 ? k : j);
+      r1[j] = (__bf16)(float)(mode == 0 ? k : 8 + j);
+      r2[j] = (__bf16)(float)(mode == 0 ? k : 16 + j);
+      r3[j] = (__bf16)(float)(mode == 0 ? k : 24 + j);
+    }
+    __bf16* base = s_w + (k >> 2) * 16 + (k & 3) * 4;
+    *(bf16x4*)(base + 0 * 1024) = lo4(r0);
+    *(bf16x4*)(base + 1 * 1024) = hi4(r0);
+    *(bf16x4*)(base + 2 * 1024) = lo4(r1);
+    *(bf16x4*)(base + 3 * 1024) = hi4(r1);
+    *(bf16x4*)(base + 4 * 1024) = lo4(r2);
+    *(bf16x4*)(base + 5 * 1024) = hi4(r2);
+    *(bf16x4*)(base + 6 * 1024) = lo4(r3);
+    *(bf16x4*)(base + 7 * 1024) = hi4(r3);
+    __syncthreads();
+    for (int i = tid; i < 8192; i += 256) {
+      out[i] = (float)s_w[i];
+    }
+    return;
+  }
+
+  for (int i = tid; i < 8192; i += 256) {
+    s_w[i] = (__bf16)(float)(mode == 2 ? (i & 255) : (i >> 8));
+  }
+  __syncthreads();
+  const __bf16* w_base = s_w + (wn * 4 + ((lane & 15) >> 2)) * 1024
+                         + (lane >> 4) * 32;
+#pragma unroll
+  for (int pr = 0; pr < 2; ++pr) {
+    const __bf16* wb = w_base + pr * 8 * 16;
+    bf16x4 lo, hi;
+    ds_tr16_pair(wb, lo, hi);
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      out[(pr * 256 + tid) * 8 + j] = (float)lo[j];
+      out[(pr * 256 + tid) * 8 + 4 + j] = (float)hi[j];
+    }
+  }
+}
+
+void tr16_diag_launch(float* out, int mode, hipStream_t stream) {
+  hipLaunchKernelGGL(tr16_diag_kernel, dim3(1), dim3(256), 0, stream,
+                     out, mode);
+  HIP_CHECK(hipGetLastError());
+}
+
+// =====================================================================
+// v3 kernels: per-lane FRAGMENT-ORDER LDS images (guide §7: fix
+// ds_read_u16 with a contiguous fragment layout).
+//
+// The tr16 probes (tools/probe_tr16_diag.py) showed the transpose-
+// read delivers only 16 distinct values per 16-lane quarter (rows
+// from the four subgroup-leader addresses, column l&3), so it cannot
+// feed full 16x16x32 fragments any cheaper than scalar reads. v3
+// instead stores each lane's fragment elements CONTIGUOUSLY: the
+// staging threads load k-strided global words and transpose u8/bf16
+// blocks in registers, so every MFMA operand becomes ONE
+// ds_read_b128 per k-step. Per wave per 64-k chunk: 18 wide reads +
+// 16 MFMAs (v1: 144 scalar reads + 16 MFMAs).
+//
+// x image (per n-tile nt): slot(l) = (l>>3)*136 + (l&7)*16 elems
+// (16 B pad per 8 lanes keeps the b128 lane banks distinct);
+// slot holds [s in 2][8] = x[k = s*32 + (l>>4)*8 + j][nt*16+(l&15)].
+// =====================================================================
+#define KC3 64
+#define DY3_MT 1088   // per-m-tile image elems: 8 groups x 136
+#define X3_NT 1088    // per-n-tile image elems
+
+__device__ __forceinline__ int frag_slot(int lane) {
+  return (lane >> 3) * 136 + (lane & 7) * 16;
+}
+
+__global__ __launch_bounds__(256)
+void conv1_wrw_v3_kernel(const __bf16* __restrict__ dy,
+                         const unsigned char* __restrict__ frames,
+                         float* __restrict__ out,
+                         float* __restrict__ b_out, int64_t K,
+                         int64_t k_per_wg, float scale) {
+  __shared__ __bf16 s_dy[2 * DY3_MT];
+  __shared__ __bf16 s_x[16 * X3_NT];
+
+  const int tid = threadIdx.x;
+  const int wave = tid / MA_WAVE;
+  const int lane = tid % MA_WAVE;
+  const int mt = wave & 1;
+  const int ng = wave >> 1;
+
+  f32x4 acc[8];
+#pragma unroll
+  for (int i = 0; i < 8; ++i) acc[i] = (f32x4)(0.0f);
+  float bias_acc = 0.0f;
+
+  const __bf16* a_base = s_dy + mt * DY3_MT + frag_slot(lane);
+  const __bf16* b_base = s_x + ng * 8 * X3_NT + frag_slot(lane);
+
+  const int64_t k_begin = (int64_t)blockIdx.x * k_per_wg;
+  const int64_t k_end = min(k_begin + k_per_wg, K);
+
+  for (int64_t kc = k_begin; kc < k_end; kc += KC3) {
+    // ---- stage dy: 64 blocks of (4 m-cols x 8 k-rows); thread
+    // loads 8x8B k-strided, transposes 4x8 bf16 in registers -------
+    if (tid < 64) {
+      int mg = tid & 7;         // m-cols mg*4..mg*4+3
+      int kg = tid >> 3;        // k-rows kg*8..kg*8+7
+      int s = kg >> 2, q = kg & 3;
+      bf16x4 ld[8];
+#pragma unroll
+      for (int i = 0; i < 8; ++i) {
+        int64_t kk = kc + kg * 8 + i;
+        if (kk < k_end) {
+          ld[i] = *(const bf16x4*)(dy + kk * CONV1_COUT + mg * 4);
+        } else {
+#pragma unroll
+          for (int j = 0; j < 4; ++j) ld[i][j] = (__bf16)0.0f;
+        }
+      }
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        int m = mg * 4 + j;
+        int l = q * 16 + (m & 15);
+        bf16x8 v;
+#pragma unroll
+        for (int i = 0; i < 8; ++i) v[i] = ld[i][j];
+        *(bf16x8*)(s_dy + (m >> 4) * DY3_MT + frag_slot(l) + s * 8) = v;
+      }
+    }
+    // ---- stage x: 512 blocks of (4 n-cols x 8 k-rows); thread
+    // loads 8 dwords (u8x4) k-strided, dequant-transposes ----------
+#pragma unroll
+    for (int blk = tid; blk < 512; blk += 256) {
+      int ngc = blk & 63;       // n-group: cols ngc*4..ngc*4+3
+      int kg = blk >> 6;        // k-rows kg*8..kg*8+7
+      int s = kg >> 2, q = kg & 3;
+      int r = ngc >> 3;               // patch row
+      int dword_off = (ngc & 7) * 4;  // byte offset in the 32B segment
+      unsigned int d[8];
+#pragma unroll
+      for (int i = 0; i < 8; ++i) {
+        int64_t kk = kc + kg * 8 + i;
+        d[i] = 0u;
+        if (kk < k_end) {
+          int64_t b = kk / CONV1_POS;
+          int pos = (int)(kk % CONV1_POS);
+          int oh = pos / CONV1_OHW, ow = pos % CONV1_OHW;
+          d[i] = *(const unsigned int*)(
+              frames
+              + ((b * 84 + (int64_t)oh * CONV1_STRIDE + r) * 84
+                 + (int64_t)ow * CONV1_STRIDE) * CONV1_CIN
+              + dword_off);
+        }
+      }
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        int n = ngc * 4 + j;
+        int l = q * 16 + (n & 15);
+        bf16x8 v;
+#pragma unroll
+        for (int i = 0; i < 8; ++i) {
+          v[i] = u8_bf16((d[i] >> (8 * j)) & 0xFF, scale);
+        }
+        *(bf16x8*)(s_x + (n >> 4) * X3_NT + frag_slot(l) + s * 8) = v;
+      }
+    }
+    __syncthreads();
+
+    // ---- MFMA: every operand is one ds_read_b128 -----------------
+#pragma unroll
+    for (int s = 0; s < 2; ++s) {
+      bf16x8 a_frag = *(const bf16x8*)(a_base + s * 8);
+      if (ng == 0) {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) bias_acc += (float)a_frag[j];
+      }
+#pragma unroll
+      for (int nt = 0; nt < 8; ++nt) {
+        bf16x8 b_frag = *(const bf16x8*)(b_base + nt * X3_NT + s * 8);
+        acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            a_frag, b_frag, acc[nt], 0, 0, 0);
+      }
+    }
+    __syncthreads();
+  }
+
+  if (ng == 0) {
+    bias_acc += __shfl_down(bias_acc, 32, 64);
+    bias_acc += __shfl_down(bias_acc, 16, 64);
+    if (lane < 16) atomicAdd(&b_out[mt * 16 + lane], bias_acc);
+  }
+#pragma unroll
+  for (int nt = 0; nt < 8; ++nt) {
+    int col = (ng * 8 + nt) * 16 + (lane & 15);
+    int row_base = mt * 16 + (lane >> 4) * 4;
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) {
+      atomicAdd(&out[(row_base + reg) * CONV1_N + col], acc[nt][reg]);
+    }
+  }
+}
+
+// fwd v3: weight staged ONCE into a fragment-order image (scalar
+// writes, amortized over the whole block); per pr-chunk the B
+// fragment is one ds_read_b128 and is shared by both m-tiles.
+// w image: addr(wn, l, pr) = wn*4624 + (l>>4)*1156 + (l&15)*72
+//          + pr*8   (4-elem pad per quarter, 8 per lane slot)
+__global__ __launch_bounds__(256)
+void conv1_fwd_v3_kernel(const unsigned char* __restrict__ frames,
+                         const __bf16* __restrict__ weight,  // [256][32]
+                         const float* __restrict__ bias,
+                         __bf16* __restrict__ out, int64_t K,
+                         float scale) {
+  __shared__ __bf16 s_x[FWD_ROWS * 32];
+  __shared__ __bf16 s_w[2 * 4624];
+
+  const int tid = threadIdx.x;
+  const int wave = tid / MA_WAVE;
+  const int lane = tid % MA_WAVE;
+  const int wm = wave & 1;
+  const int wn = wave >> 1;
+
+  {
+    int k = tid;  // one weight row per thread
+    int pr = k >> 5, q = (k >> 3) & 3, i = k & 7;
+    bf16x8 r0 = ((const bf16x8*)(weight + k * 32))[0];
+    bf16x8 r1 = ((const bf16x8*)(weight + k * 32))[1];
+    bf16x8 r2 = ((const bf16x8*)(weight + k * 32))[2];
+    bf16x8 r3 = ((const bf16x8*)(weight + k * 32))[3];
+#pragma unroll
+    for (int n16 = 0; n16 < 16; ++n16) {
+      int l = q * 16 + n16;
+      s_w[(l >> 4) * 1156 + (l & 15) * 72 + pr * 8 + i] =
+          n16 < 8 ? r0[n16] : r1[n16 - 8];
+      s_w[4624 + (l >> 4) * 1156 + (l & 15) * 72 + pr * 8 + i] =
+          n16 < 8 ? r2[n16] : r3[n16 - 8];
+    }
+  }
+
+  const int64_t row0 = (int64_t)blockIdx.x * FWD_ROWS;
+  f32x4 acc[2];
+  acc[0] = (f32x4)(0.0f);
+  acc[1] = (f32x4)(0.0f);
+  const __bf16* w_base = s_w + wn * 4624 + (lane >> 4) * 1156
+                         + (lane & 15) * 72;
+  __syncthreads();
+
+  for (int pr = 0; pr < CONV1_KSZ; ++pr) {
+    {
+      int krow = tid >> 2;
+      int quarter = tid & 3;
+      int64_t kk = row0 + krow;
+      bf16x8 q8;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) q8[j] = (__bf16)0.0f;
+      if (kk < K) {
+        int64_t b = kk / CONV1_POS;
+        int pos = (int)(kk % CONV1_POS);
+        int oh = pos / CONV1_OHW, ow = pos % CONV1_OHW;
+        const unsigned char* src =
+            frames
+            + ((b * 84 + (int64_t)oh * CONV1_STRIDE + pr) * 84
+               + (int64_t)ow * CONV1_STRIDE) * CONV1_CIN
+            + quarter * 8;
+        uint2 raw = *(const uint2*)src;
+        unsigned int words[2] = {raw.x, raw.y};
+#pragma unroll
+        for (int w = 0; w < 2; ++w) {
+#pragma unroll
+          for (int j = 0; j < 4; ++j) {
+            q8[w * 4 + j] = u8_bf16((words[w] >> (8 * j)) & 0xFF,
+                                    scale);
+          }
+        }
+      }
+      *(bf16x8*)(s_x + krow * 32 + quarter * 8) = q8;
+    }
+    __syncthreads();
+
+    bf16x8 b_frag = *(const bf16x8*)(w_base + pr * 8);
 #pragma unroll
     for (int mt2 = 0; mt2 < 2; ++mt2) {
       int m = wm * 32 + mt2 * 16 + (lane & 15);

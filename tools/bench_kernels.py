@@ -149,6 +149,25 @@ def main():
 
     results["nstep_64x4096_n3_torch_ms"] = timeit(nstep_torch)
 
+    # --- conv1 stem kernels: v1 vs v3 A/B at 1/5 bench K ------------
+    Bc = 8192
+    Kc = Bc * 400
+    frames_c = t.randint(0, 256, (Bc, 84, 84, 4), dtype=t.uint8,
+                         device=dev)
+    gy = (t.randn(Kc, 32, device=dev) * 0.1).to(t.bfloat16).contiguous()
+    wgt = (t.randn(256, 32, device=dev) * 0.1).to(t.bfloat16).contiguous()
+    bias_c = t.zeros(32, device=dev)
+    for v in ("1", "3"):
+        os.environ["MACHIN_CONV1_V"] = v
+        results[f"conv1_fwd_v{v}_K3.3M_ms"] = timeit(
+            lambda: ext.conv1_fwd(frames_c, wgt, bias_c, 1.0 / 255.0),
+            iters=30,
+        )
+        results[f"conv1_wrw_v{v}_K3.3M_ms"] = timeit(
+            lambda: ext.conv1_wrw(gy, frames_c, 1.0 / 255.0), iters=30
+        )
+    os.environ.pop("MACHIN_CONV1_V", None)
+
     os.makedirs("gpurun_out", exist_ok=True)
     with open("gpurun_out/kernel_bench.json", "w") as f:
         json.dump(results, f, indent=2)

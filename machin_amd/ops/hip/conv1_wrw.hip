@@ -399,7 +399,8 @@ void conv1_fwd_launch(const unsigned char* frames, const void* weight,
                        (const __bf16*)weight, bias, (__bf16*)out, K,
                        scale);
   } else {
-    hipLaunchKernelGGL(conv1_fwd_v3_kernel, dim3((unsigned)grid),
+    int64_t grid3 = (K + 127) / 128;  // FWD3_ROWS
+    hipLaunchKernelGGL(conv1_fwd_v3_kernel, dim3((unsigned)grid3),
                        dim3(256), 0, stream, frames,
                        (const __bf16*)weight, bias, (__bf16*)out, K,
                        scale);
@@ -962,74 +963,92 @@ void conv1_wrw_v3_kernel(const __bf16* __restrict__ dy,
   const int64_t k_begin = (int64_t)blockIdx.x * k_per_wg;
   const int64_t k_end = min(k_begin + k_per_wg, K);
 
-  for (int64_t kc = k_begin; kc < k_end; kc += KC3) {
-    // ---- stage dy: 64 blocks of (4 m-cols x 8 k-rows); thread
-    // loads 8x8B k-strided, transposes 4x8 bf16 in registers -------
-    if (tid < 64) {
-      int mg = tid & 7;         // m-cols mg*4..mg*4+3
-      int kg = tid >> 3;        // k-rows kg*8..kg*8+7
-      int s = kg >> 2, q = kg & 3;
-      bf16x4 ld[8];
+  // register prefetch: global loads for chunk kc+KC3 are issued while
+  // the MFMAs for chunk kc run (PMC showed 35-48% SQ_WAIT_ANY =
+  // waves parked on global latency with a serial stage->barrier->
+  // MFMA loop)
+  const int x_ngc = tid & 63, x_kg = tid >> 6;  // x block 1 of 2
+  const int x_ngc2 = x_ngc, x_kg2 = x_kg + 4;   // x block 2
+  const int dy_mg = tid & 7, dy_kg = tid >> 3;  // dy block (tid<64)
+
+  auto load_x = [&](int64_t kc, int ngc, int kg, unsigned int* d) {
+    int r = ngc >> 3;
+    int dword_off = (ngc & 7) * 4;
+#pragma unroll
+    for (int i = 0; i < 8; ++i) {
+      int64_t kk = kc + kg * 8 + i;
+      d[i] = 0u;
+      if (kk < k_end) {
+        int64_t b = kk / CONV1_POS;
+        int pos = (int)(kk % CONV1_POS);
+        int oh = pos / CONV1_OHW, ow = pos % CONV1_OHW;
+        d[i] = *(const unsigned int*)(
+            frames
+            + ((b * 84 + (int64_t)oh * CONV1_STRIDE + r) * 84
+               + (int64_t)ow * CONV1_STRIDE) * CONV1_CIN
+            + dword_off);
+      }
+    }
+  };
+  auto load_dy = [&](int64_t kc, bf16x4* ld) {
+#pragma unroll
+    for (int i = 0; i < 8; ++i) {
+      int64_t kk = kc + dy_kg * 8 + i;
+      if (kk < k_end) {
+        ld[i] = *(const bf16x4*)(dy + kk * CONV1_COUT + dy_mg * 4);
+      } else {
+#pragma unroll
+        for (int j = 0; j < 4; ++j) ld[i][j] = (__bf16)0.0f;
+      }
+    }
+  };
+  auto store_x = [&](int ngc, int kg, const unsigned int* d) {
+    int s = kg >> 2, q = kg & 3;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      int n = ngc * 4 + j;
+      int l = q * 16 + (n & 15);
+      bf16x8 v;
 #pragma unroll
       for (int i = 0; i < 8; ++i) {
-        int64_t kk = kc + kg * 8 + i;
-        if (kk < k_end) {
-          ld[i] = *(const bf16x4*)(dy + kk * CONV1_COUT + mg * 4);
-        } else {
-#pragma unroll
-          for (int j = 0; j < 4; ++j) ld[i][j] = (__bf16)0.0f;
-        }
+        v[i] = u8_bf16((d[i] >> (8 * j)) & 0xFF, scale);
       }
+      *(bf16x8*)(s_x + (n >> 4) * X3_NT + frag_slot(l) + s * 8) = v;
+    }
+  };
+
+  unsigned int d1[8], d2[8];
+  bf16x4 ldy[8];
+  load_x(k_begin, x_ngc, x_kg, d1);
+  load_x(k_begin, x_ngc2, x_kg2, d2);
+  if (tid < 64) load_dy(k_begin, ldy);
+
+  for (int64_t kc = k_begin; kc < k_end; kc += KC3) {
+    // write the prefetched chunk into the fragment images
+    if (tid < 64) {
+      int s = dy_kg >> 2, q = dy_kg & 3;
 #pragma unroll
       for (int j = 0; j < 4; ++j) {
-        int m = mg * 4 + j;
+        int m = dy_mg * 4 + j;
         int l = q * 16 + (m & 15);
         bf16x8 v;
 #pragma unroll
-        for (int i = 0; i < 8; ++i) v[i] = ld[i][j];
-        *(bf16x8*)(s_dy + (m >> 4) * DY3_MT + frag_slot(l) + s * 8) = v;
+        for (int i = 0; i < 8; ++i) v[i] = ldy[i][j];
+        *(bf16x8*)(s_dy + (m >> 4) * DY3_MT + frag_slot(l)
+                   + s * 8) = v;
       }
     }
-    // ---- stage x: 512 blocks of (4 n-cols x 8 k-rows); thread
-    // loads 8 dwords (u8x4) k-strided, dequant-transposes ----------
-#pragma unroll
-    for (int blk = tid; blk < 512; blk += 256) {
-      int ngc = blk & 63;       // n-group: cols ngc*4..ngc*4+3
-      int kg = blk >> 6;        // k-rows kg*8..kg*8+7
-      int s = kg >> 2, q = kg & 3;
-      int r = ngc >> 3;               // patch row
-      int dword_off = (ngc & 7) * 4;  // byte offset in the 32B segment
-      unsigned int d[8];
-#pragma unroll
-      for (int i = 0; i < 8; ++i) {
-        int64_t kk = kc + kg * 8 + i;
-        d[i] = 0u;
-        if (kk < k_end) {
-          int64_t b = kk / CONV1_POS;
-          int pos = (int)(kk % CONV1_POS);
-          int oh = pos / CONV1_OHW, ow = pos % CONV1_OHW;
-          d[i] = *(const unsigned int*)(
-              frames
-              + ((b * 84 + (int64_t)oh * CONV1_STRIDE + r) * 84
-                 + (int64_t)ow * CONV1_STRIDE) * CONV1_CIN
-              + dword_off);
-        }
-      }
-#pragma unroll
-      for (int j = 0; j < 4; ++j) {
-        int n = ngc * 4 + j;
-        int l = q * 16 + (n & 15);
-        bf16x8 v;
-#pragma unroll
-        for (int i = 0; i < 8; ++i) {
-          v[i] = u8_bf16((d[i] >> (8 * j)) & 0xFF, scale);
-        }
-        *(bf16x8*)(s_x + (n >> 4) * X3_NT + frag_slot(l) + s * 8) = v;
-      }
-    }
+    store_x(x_ngc, x_kg, d1);
+    store_x(x_ngc2, x_kg2, d2);
     __syncthreads();
 
-    // ---- MFMA: every operand is one ds_read_b128 -----------------
+    // issue next chunk's global loads, then MFMA over this chunk —
+    // the loads complete under the MFMAs
+    if (kc + KC3 < k_end) {
+      load_x(kc + KC3, x_ngc, x_kg, d1);
+      load_x(kc + KC3, x_ngc2, x_kg2, d2);
+      if (tid < 64) load_dy(kc + KC3, ldy);
+    }
 #pragma unroll
     for (int s = 0; s < 2; ++s) {
       bf16x8 a_frag = *(const bf16x8*)(a_base + s * 8);
@@ -1063,25 +1082,28 @@ void conv1_wrw_v3_kernel(const __bf16* __restrict__ dy,
   }
 }
 
-// fwd v3: weight staged ONCE into a fragment-order image (scalar
-// writes, amortized over the whole block); per pr-chunk the B
-// fragment is one ds_read_b128 and is shared by both m-tiles.
+// fwd v3: 128 output rows per workgroup (16 MFMAs between barrier
+// pairs instead of 8) with register-prefetched frame loads — the PMC
+// run showed fwd 71% SQ_WAIT_ANY (waves parked on global latency in
+// a serial stage->barrier->MFMA loop). Weight staged ONCE into a
+// fragment-order image; per pr-chunk each B fragment is one
+// ds_read_b128 shared across both m-tiles of the wave.
 // w image: addr(wn, l, pr) = wn*4624 + (l>>4)*1156 + (l&15)*72
 //          + pr*8   (4-elem pad per quarter, 8 per lane slot)
+#define FWD3_ROWS 128
+
 __global__ __launch_bounds__(256)
 void conv1_fwd_v3_kernel(const unsigned char* __restrict__ frames,
                          const __bf16* __restrict__ weight,  // [256][32]
                          const float* __restrict__ bias,
                          __bf16* __restrict__ out, int64_t K,
                          float scale) {
-  __shared__ __bf16 s_x[FWD_ROWS * 32];
+  __shared__ __bf16 s_x[FWD3_ROWS * 32];
   __shared__ __bf16 s_w[2 * 4624];
 
   const int tid = threadIdx.x;
-  const int wave = tid / MA_WAVE;
+  const int wave = tid / MA_WAVE;  // wave w owns rows w*32..w*32+31
   const int lane = tid % MA_WAVE;
-  const int wm = wave & 1;
-  const int wn = wave >> 1;
 
   {
     int k = tid;  // one weight row per thread
@@ -1100,68 +1122,83 @@ void conv1_fwd_v3_kernel(const unsigned char* __restrict__ frames,
     }
   }
 
-  const int64_t row0 = (int64_t)blockIdx.x * FWD_ROWS;
-  f32x4 acc[2];
-  acc[0] = (f32x4)(0.0f);
-  acc[1] = (f32x4)(0.0f);
-  const __bf16* w_base = s_w + wn * 4624 + (lane >> 4) * 1156
-                         + (lane & 15) * 72;
-  __syncthreads();
+  const int64_t row0 = (int64_t)blockIdx.x * FWD3_ROWS;
+  f32x4 acc[2][2];  // [m-tile][n-tile]
+#pragma unroll
+  for (int a = 0; a < 2; ++a) {
+    acc[a][0] = (f32x4)(0.0f);
+    acc[a][1] = (f32x4)(0.0f);
+  }
+  const __bf16* w_base = s_w + (lane >> 4) * 1156 + (lane & 15) * 72;
+
+  // staging: 2 threads per row, 16 B each
+  const int st_row = tid >> 1, st_half = tid & 1;
+  const int64_t st_kk = row0 + st_row;
+  const unsigned char* st_src = nullptr;
+  if (st_kk < K) {
+    int64_t b = st_kk / CONV1_POS;
+    int pos = (int)(st_kk % CONV1_POS);
+    int oh = pos / CONV1_OHW, ow = pos % CONV1_OHW;
+    st_src = frames
+             + ((b * 84 + (int64_t)oh * CONV1_STRIDE) * 84
+                + (int64_t)ow * CONV1_STRIDE) * CONV1_CIN
+             + st_half * 16;
+  }
+
+  uint4 raw = {0u, 0u, 0u, 0u};
+  if (st_src != nullptr) raw = *(const uint4*)st_src;  // pr=0
+  __syncthreads();  // weight image ready
 
   for (int pr = 0; pr < CONV1_KSZ; ++pr) {
     {
-      int krow = tid >> 2;
-      int quarter = tid & 3;
-      int64_t kk = row0 + krow;
-      bf16x8 q8;
+      unsigned int words[4] = {raw.x, raw.y, raw.z, raw.w};
+      bf16x8 q0, q1;
 #pragma unroll
-      for (int j = 0; j < 8; ++j) q8[j] = (__bf16)0.0f;
-      if (kk < K) {
-        int64_t b = kk / CONV1_POS;
-        int pos = (int)(kk % CONV1_POS);
-        int oh = pos / CONV1_OHW, ow = pos % CONV1_OHW;
-        const unsigned char* src =
-            frames
-            + ((b * 84 + (int64_t)oh * CONV1_STRIDE + pr) * 84
-               + (int64_t)ow * CONV1_STRIDE) * CONV1_CIN
-            + quarter * 8;
-        uint2 raw = *(const uint2*)src;
-        unsigned int words[2] = {raw.x, raw.y};
+      for (int w = 0; w < 2; ++w) {
 #pragma unroll
-        for (int w = 0; w < 2; ++w) {
-#pragma unroll
-          for (int j = 0; j < 4; ++j) {
-            q8[w * 4 + j] = u8_bf16((words[w] >> (8 * j)) & 0xFF,
-                                    scale);
-          }
+        for (int j = 0; j < 4; ++j) {
+          q0[w * 4 + j] = u8_bf16((words[w] >> (8 * j)) & 0xFF, scale);
+          q1[w * 4 + j] =
+              u8_bf16((words[w + 2] >> (8 * j)) & 0xFF, scale);
         }
       }
-      *(bf16x8*)(s_x + krow * 32 + quarter * 8) = q8;
+      *(bf16x8*)(s_x + st_row * 32 + st_half * 16) = q0;
+      *(bf16x8*)(s_x + st_row * 32 + st_half * 16 + 8) = q1;
     }
     __syncthreads();
-
-    bf16x8 b_frag = *(const bf16x8*)(w_base + pr * 8);
+    // prefetch next patch row while the MFMAs below run
+    if (pr + 1 < CONV1_KSZ && st_src != nullptr) {
+      raw = *(const uint4*)(st_src + (int64_t)(pr + 1) * 84
+                            * CONV1_CIN);
+    }
+    bf16x8 b0 = *(const bf16x8*)(w_base + pr * 8);
+    bf16x8 b1 = *(const bf16x8*)(w_base + 4624 + pr * 8);
 #pragma unroll
     for (int mt2 = 0; mt2 < 2; ++mt2) {
-      int m = wm * 32 + mt2 * 16 + (lane & 15);
+      int m = wave * 32 + mt2 * 16 + (lane & 15);
       int k0 = (lane >> 4) * 8;
       bf16x8 a_frag = *(bf16x8*)(s_x + m * 32 + k0);
-      acc[mt2] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-          a_frag, b_frag, acc[mt2], 0, 0, 0);
+      acc[mt2][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+          a_frag, b0, acc[mt2][0], 0, 0, 0);
+      acc[mt2][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+          a_frag, b1, acc[mt2][1], 0, 0, 0);
     }
     __syncthreads();
   }
 
 #pragma unroll
   for (int mt2 = 0; mt2 < 2; ++mt2) {
-    int col = wn * 16 + (lane & 15);
-    float b = bias != nullptr ? bias[col] : 0.0f;
-    int row_base = wm * 32 + mt2 * 16 + (lane >> 4) * 4;
 #pragma unroll
-    for (int reg = 0; reg < 4; ++reg) {
-      int64_t kk = row0 + row_base + reg;
-      if (kk < K) {
-        out[kk * 32 + col] = (__bf16)(acc[mt2][reg] + b);
+    for (int nt2 = 0; nt2 < 2; ++nt2) {
+      int col = nt2 * 16 + (lane & 15);
+      float b = bias != nullptr ? bias[col] : 0.0f;
+      int row_base = wave * 32 + mt2 * 16 + (lane >> 4) * 4;
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        int64_t kk = row0 + row_base + reg;
+        if (kk < K) {
+          out[kk * 32 + col] = (__bf16)(acc[mt2][nt2][reg] + b);
+        }
       }
     }
   }

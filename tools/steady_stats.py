@@ -11,15 +11,25 @@ import sys
 
 def main():
     path = sys.argv[1]
-    frac = float(sys.argv[2]) if len(sys.argv) > 2 else 0.25
-    out_md = sys.argv[3] if len(sys.argv) > 3 else None
+    anchor = sys.argv[2] if len(sys.argv) > 2 else "conv1_fwd"
+    n_steps = int(sys.argv[3]) if len(sys.argv) > 3 else 30
+    out_md = sys.argv[4] if len(sys.argv) > 4 else None
     rows = list(csv.DictReader(open(path)))
     skey = next(k for k in rows[0] if "Start" in k)
     ekey = next(k for k in rows[0] if "End" in k)
     nkey = next(k for k in rows[0] if "Kernel_Name" in k or "Name" in k)
-    t0 = min(int(r[skey]) for r in rows)
+    # anchor the steady window on the last n_steps dispatches of a
+    # once-per-step kernel: immune to autotune kernels interleaved
+    # anywhere else in the trace
+    anchors = sorted(
+        int(r[skey]) for r in rows if anchor in r[nkey]
+    )
+    if len(anchors) < n_steps + 1:
+        raise SystemExit(f"only {len(anchors)} {anchor!r} dispatches")
+    cut = anchors[-n_steps]
     t1 = max(int(r[ekey]) for r in rows)
-    cut = t1 - (t1 - t0) * frac
+    t0 = cut
+    frac = n_steps
     agg = collections.defaultdict(lambda: [0, 0.0])
     total = 0.0
     for r in rows:
@@ -31,8 +41,9 @@ def main():
         agg[name][1] += dur
         total += dur
     lines = [
-        f"# Steady-state kernel profile (final {frac:.0%} of "
-        f"{(t1 - t0) / 1e9:.1f}s trace; autotune window excluded)",
+        f"# Steady-state kernel profile: window = last {n_steps} "
+        f"dispatches of '{anchor}' ({(t1 - t0) / 1e6:.0f} ms; MIOpen "
+        f"autotune/find phases outside the window are EXCLUDED)",
         "",
         "| % | total ms | calls | avg ms | kernel |",
         "|---|---|---|---|---|",

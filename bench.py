@@ -24,6 +24,7 @@ Scaling is WEAK: per-GPU work (unroll x env_batch) is fixed as N grows.
 import argparse
 import json
 import os
+import sys
 import threading
 import time
 
@@ -169,30 +170,47 @@ class RingPipeline:
         while not self._stop_ingest:
             self.ingest_once()
 
-    def sample(self):
+    def sample(self, out=None):
         """Gather a t-major training batch from the HBM pool: one
         index_select plus ONE strided copy straight into the
         channels-last layout the conv stem wants (the round-1 chain
         permute -> reshape -> to(channels_last) made three full passes
-        over the 1.2 GB frame batch)."""
+        over the 1.2 GB frame batch). With ``out`` the gather writes
+        into preallocated static buffers (hipGraph replay mode)."""
         B, T = self.env_batch, self.unroll
         seg = t.randint(0, max(self.filled, 1), (B,), device=self.device)
         p = self.pool
         fr = p["frames"].index_select(0, seg)  # [B, T, C, H, W]
         C, H, W = fr.shape[2:]
-        frames = t.empty(
-            (T * B, C, H, W), dtype=fr.dtype, device=self.device,
-            memory_format=t.channels_last,
-        )
-        frames.view(T, B, C, H, W).copy_(fr.transpose(0, 1))
+        if out is None:
+            frames = t.empty(
+                (T * B, C, H, W), dtype=fr.dtype, device=self.device,
+                memory_format=t.channels_last,
+            )
+            out = {"frames": frames}
+            out["frames"].view(T, B, C, H, W).copy_(fr.transpose(0, 1))
+            for k in ("actions", "behavior_logp", "rewards",
+                      "terminals"):
+                out[k] = p[k].index_select(0, seg).t().contiguous()
+            return out
+        out["frames"].view(T, B, C, H, W).copy_(fr.transpose(0, 1))
+        for k in ("actions", "behavior_logp", "rewards", "terminals"):
+            out[k].copy_(p[k].index_select(0, seg).t())
+        return out
+
+    def make_static_batch(self):
+        B, T = self.env_batch, self.unroll
+        C, H, W = self.pool["frames"].shape[2:]
         return {
-            "frames": frames,
-            "actions": p["actions"].index_select(0, seg).t().contiguous(),
-            "behavior_logp": p["behavior_logp"].index_select(0, seg)
-            .t().contiguous(),
-            "rewards": p["rewards"].index_select(0, seg).t().contiguous(),
-            "terminals": p["terminals"].index_select(0, seg)
-            .t().contiguous(),
+            "frames": t.empty(
+                (T * B, C, H, W), dtype=t.uint8, device=self.device,
+                memory_format=t.channels_last,
+            ),
+            "actions": t.empty((T, B), dtype=t.long,
+                               device=self.device),
+            "behavior_logp": t.empty((T, B), device=self.device),
+            "rewards": t.empty((T, B), device=self.device),
+            "terminals": t.empty((T, B), device=self.device),
         }
 
     def reset_env_steps(self):
@@ -323,6 +341,12 @@ class ImpalaLearnerBench:
         """One learner step; returns the detached loss TENSOR (no
         host synchronization on the hot path)."""
         if self.pipeline is not None:
+            if self._graph is not None:
+                # gather straight into the graph's static buffers,
+                # then one replay instead of ~300 launches
+                self.pipeline.sample(out=self._static_in)
+                self._graph.replay()
+                return self._static_loss
             return self._step_body(self.pipeline.sample())
         data = self.pool[self._pool_i]
         self._pool_i = (self._pool_i + 1) % len(self.pool)
@@ -392,15 +416,19 @@ class ImpalaLearnerBench:
         """Capture the whole learner step in a hipGraph: the Nature
         CNN is small, so kernel-launch overhead is a real cost at low
         batch — one graph replay replaces ~300 launches. Single-GPU
-        learner-only mode (RCCL collectives stay outside graphs)."""
+        only (RCCL collectives stay outside graphs); works in both
+        learner-only and pipeline mode (the pool gather runs eagerly
+        into the graph's static input buffers each step)."""
         if self.reducer is not None:
             raise RuntimeError("graph capture is single-GPU only")
         if self.pipeline is not None:
-            raise RuntimeError("graph capture requires --learner-only")
-        data = self.pool[0]
-        self._static_in = {
-            k: v.clone() for k, v in data.items()
-        }
+            self._static_in = self.pipeline.make_static_batch()
+            self.pipeline.sample(out=self._static_in)
+        else:
+            data = self.pool[0]
+            self._static_in = {
+                k: v.clone() for k, v in data.items()
+            }
         side = t.cuda.Stream()
         side.wait_stream(t.cuda.current_stream())
         with t.cuda.stream(side):
@@ -428,8 +456,10 @@ def main():
                         help="skip the actor farm/ring; measure the "
                              "bare learner loop from resident pools")
     parser.add_argument("--graph", action="store_true",
-                        help="capture the learner step in a hipGraph "
-                             "(learner-only mode)")
+                        help="force hipGraph capture of the learner "
+                             "step (default: on for single-GPU runs)")
+    parser.add_argument("--no-graph", action="store_true",
+                        help="disable hipGraph capture")
     parser.add_argument("--bucket-mb", type=float, default=32.0,
                         help="GradReducer bucket size (MiB)")
     parser.add_argument("--ddp-reduction", default="all_reduce",
@@ -468,9 +498,6 @@ def main():
         os.environ.get("MACHIN_MIOPEN_BENCHMARK", "1") == "1"
     )
 
-    if args.graph and not args.learner_only:
-        args.learner_only = True
-
     pipeline = None
     if not args.learner_only:
         n_actors = args.actors
@@ -483,13 +510,14 @@ def main():
             pool_segments=args.pool_segments,
         )
 
+    use_graph = (args.graph or not args.no_graph) and not distributed
     bench = ImpalaLearnerBench(
         device=device,
         unroll=args.unroll,
         env_batch=args.env_batch,
         action_num=args.actions,
         distributed=distributed,
-        capturable=args.graph and not distributed,
+        capturable=use_graph,
         fused_stem=not args.no_fused_stem,
         bucket_mb=args.bucket_mb,
         ddp_reduction=args.ddp_reduction,
@@ -498,8 +526,14 @@ def main():
 
     if pipeline is not None:
         pipeline.start()
-    if args.graph and not distributed:
-        bench.capture_graph()
+    if use_graph:
+        try:
+            bench.capture_graph()
+            print("hipGraph capture: ON", file=sys.stderr)
+        except Exception as e:  # noqa: BLE001 - fall back to eager
+            print(f"hipGraph capture failed ({e}); eager mode",
+                  file=sys.stderr)
+            bench._graph = None
     for _ in range(args.warmup):
         bench.step()
     if distributed:

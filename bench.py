@@ -123,6 +123,7 @@ class RingPipeline:
         self.filled = 0
         self.write_pos = 0
         self._stop_ingest = False
+        self._pause_ingest = False
         self._ingester = None
 
     def start(self):
@@ -168,6 +169,9 @@ class RingPipeline:
         # background staging, overlapped with training (host memcpys
         # release the GIL)
         while not self._stop_ingest:
+            if self._pause_ingest:
+                time.sleep(0.005)
+                continue
             self.ingest_once()
 
     def sample(self, out=None):
@@ -422,6 +426,10 @@ class ImpalaLearnerBench:
         if self.reducer is not None:
             raise RuntimeError("graph capture is single-GPU only")
         if self.pipeline is not None:
+            # concurrent side-stream ingest breaks stream capture
+            self.pipeline._pause_ingest = True
+            time.sleep(0.1)
+            t.cuda.synchronize()
             self._static_in = self.pipeline.make_static_batch()
             self.pipeline.sample(out=self._static_in)
         else:
@@ -438,6 +446,8 @@ class ImpalaLearnerBench:
         self._graph = t.cuda.CUDAGraph()
         with t.cuda.graph(self._graph):
             self._static_loss = self._step_body(self._static_in)
+        if self.pipeline is not None:
+            self.pipeline._pause_ingest = False
 
 
 def main():
@@ -510,7 +520,10 @@ def main():
             pool_segments=args.pool_segments,
         )
 
-    use_graph = (args.graph or not args.no_graph) and not distributed
+    # measured: graph replay is NET-NEGATIVE here (3.70 vs 3.86 M
+    # samples/s learner-only) — the step is GPU-busy-bound, not
+    # launch-gap-bound — so capture stays opt-in
+    use_graph = args.graph and not distributed and not args.no_graph
     bench = ImpalaLearnerBench(
         device=device,
         unroll=args.unroll,

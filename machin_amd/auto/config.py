@@ -22,7 +22,7 @@ def get_available_algorithms():
 
 
 def get_available_environments():
-    return ["CartPole-v1", "Pendulum-v1"]
+    return ["CartPole-v1", "Pendulum-v1", "PixelCatch-v0"]
 
 
 def _as_dict(config) -> Dict[str, Any]:
@@ -74,6 +74,10 @@ _DEFAULT_MODELS = {
 _ENV_DIMS = {
     "CartPole-v1": {"disc": {"state_dim": 4, "action_num": 2}},
     "Pendulum-v1": {"cont": {"state_dim": 3, "action_dim": 1}},
+    # pixel env flattened for the MLP zoo (CNN models are wired
+    # manually; the auto datasets flatten observations)
+    "PixelCatch-v0": {"disc": {"state_dim": 4 * 84 * 84,
+                               "action_num": 3}},
 }
 
 

@@ -158,7 +158,7 @@ def generate_env_config(environment: str, config=None):
 def make_dataset(frame, config) -> RLDataset:
     env = make(_env_name(config))
     base = _env_name(config).split("-")[0].lower()
-    if base == "cartpole":
+    if base in ("cartpole", "pixelcatch"):
         return RLGymDiscActDataset(frame, env)
     return RLGymContActDataset(frame, env)
 

@@ -112,3 +112,29 @@ class TestCLI:
         )
         assert r.returncode == 0
         assert "CartPole-v1" in r.stdout
+
+
+class TestPixelCatchAutoEnv:
+    def test_pixelcatch_config_and_one_update(self):
+        """PixelCatch is launchable through the auto layer (flattened
+        pixels on the MLP zoo; CNN models are wired manually)."""
+        from machin_amd.auto.config import (
+            generate_algorithm_config,
+            generate_env_config,
+            get_available_environments,
+            init_algorithm_from_config,
+        )
+        from machin_amd.auto.envs.classic_control import make_dataset
+
+        assert "PixelCatch-v0" in get_available_environments()
+        cfg = generate_env_config("PixelCatch-v0")
+        cfg = generate_algorithm_config("DQN", cfg)
+        cfg["frame_config"]["batch_size"] = 8
+        frame = init_algorithm_from_config(cfg)
+        ds = make_dataset(frame, cfg.data)
+        episode = next(iter(ds)).observations[0]
+        assert episode[0]["state"]["state"].shape == (1, 4 * 84 * 84)
+        frame.store_episode(episode)
+        import numpy as np
+
+        assert np.isfinite(frame.update())

@@ -403,12 +403,13 @@ class ImpalaLearnerBench:
                 frames = data["frames"].to(self.dtype).mul_(1.0 / 255.0)
             with t.autocast(device_type="cuda", dtype=self.dtype):
                 logits, values = self.model(frames)
-        logits = logits.float().view(T, B, -1)
         values = values.float().view(T, B)
-        log_pi = t.log_softmax(logits, dim=-1)
-        taken_logp = log_pi.gather(
-            -1, data["actions"].unsqueeze(-1)
-        ).squeeze(-1)
+        # fused policy head: ONE kernel for log_softmax+gather+entropy
+        # (and one analytic backward) instead of the eager chain
+        tl_flat, ent_flat = self.ops.categorical_policy_head(
+            logits.reshape(T * B, -1), data["actions"].reshape(-1)
+        )
+        taken_logp = tl_flat.view(T, B)
 
         with t.no_grad():
             bootstrap = values[-1].detach()
@@ -424,7 +425,7 @@ class ImpalaLearnerBench:
 
         pg_loss = -(pg_adv * taken_logp).sum() / B
         value_loss = 0.5 * ((vs - values) ** 2).sum() / B
-        entropy = -(log_pi.exp() * log_pi).sum(dim=-1).sum() / B
+        entropy = ent_flat.sum() / B
         loss = (
             pg_loss
             + self.value_weight * value_loss

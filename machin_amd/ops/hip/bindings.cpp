@@ -43,6 +43,10 @@ void ou_update_launch(float*, int64_t, float, float, float, float, uint64_t,
                       uint64_t, hipStream_t);
 void u8_to_bf16_scale_launch(const unsigned char*, void*, int64_t, float,
                              hipStream_t);
+void pg_head_fwd_launch(const void*, const int64_t*, float*, float*,
+                        int64_t, int, hipStream_t);
+void pg_head_bwd_launch(const void*, const int64_t*, const float*,
+                        const float*, void*, int64_t, int, hipStream_t);
 void conv1_wrw_launch(const void*, const unsigned char*, float*, float*,
                       float*, int64_t, float, hipStream_t);
 void mfma_probe_launch(const void*, const void*, float*, hipStream_t);
@@ -274,6 +278,36 @@ void ou_update_(Tensor x, double mu, double theta, double sigma, double dt,
                    (uint64_t)offset, current_stream());
 }
 
+std::vector<Tensor> pg_head_fwd(Tensor logits, Tensor actions) {
+  TORCH_CHECK(logits.is_cuda() && logits.scalar_type() == at::kBFloat16 &&
+                  logits.is_contiguous(),
+              "logits must be contiguous bf16 CUDA [N, A]");
+  TORCH_CHECK(actions.scalar_type() == at::kLong && actions.is_contiguous(),
+              "actions must be contiguous int64");
+  int64_t N = logits.size(0), A = logits.size(1);
+  TORCH_CHECK(A <= 32, "pg_head supports at most 32 actions");
+  const at::hip::OptionalHIPGuardMasqueradingAsCUDA guard(logits.device());
+  Tensor tl = at::empty({N}, logits.options().dtype(at::kFloat));
+  Tensor ent = at::empty({N}, logits.options().dtype(at::kFloat));
+  pg_head_fwd_launch(logits.data_ptr(), actions.data_ptr<int64_t>(),
+                     tl.data_ptr<float>(), ent.data_ptr<float>(), N,
+                     (int)A, current_stream());
+  return {tl, ent};
+}
+
+Tensor pg_head_bwd(Tensor logits, Tensor actions, Tensor g_taken,
+                   Tensor g_ent) {
+  int64_t N = logits.size(0), A = logits.size(1);
+  check_f32_cuda(g_taken, "g_taken");
+  check_f32_cuda(g_ent, "g_ent");
+  const at::hip::OptionalHIPGuardMasqueradingAsCUDA guard(logits.device());
+  Tensor d = at::empty_like(logits);
+  pg_head_bwd_launch(logits.data_ptr(), actions.data_ptr<int64_t>(),
+                     g_taken.data_ptr<float>(), g_ent.data_ptr<float>(),
+                     d.data_ptr(), N, (int)A, current_stream());
+  return d;
+}
+
 Tensor u8_to_bf16_scale(Tensor in, double scale) {
   TORCH_CHECK(in.is_cuda() && in.scalar_type() == at::kByte &&
                   in.is_contiguous(),
@@ -397,6 +431,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("normal_noise_", &normal_noise_);
   m.def("ou_update_", &ou_update_);
   m.def("u8_to_bf16_scale", &u8_to_bf16_scale);
+  m.def("pg_head_fwd", &pg_head_fwd);
+  m.def("pg_head_bwd", &pg_head_bwd);
   m.def("conv1_wrw", &conv1_wrw);
   m.def("mfma_probe", &mfma_probe);
   m.def("tr16_probe", &tr16_probe);

@@ -69,3 +69,105 @@ void u8_to_bf16_scale_launch(const unsigned char* in, void* out, int64_t n,
                        scale);
   }
 }
+
+// ---------------------------------------------------------------------
+// fused categorical policy head (IMPALA/A2C loss block).
+//
+// Replaces the eager chain logits.float() -> log_softmax -> gather ->
+// exp/mul/sum entropy (~6 kernels forward, ~8 backward over the
+// [T*B, A] logits) with ONE kernel each way. A is small (Atari: 6),
+// so the whole row lives in registers; one thread per row.
+//
+// forward:  taken_logp[i] = log_softmax(logits[i])[a_i]
+//           entropy[i]    = -sum_a p log p
+// backward: dlogits[i,a] = g_taken[i] * (1[a==a_i] - p_a)
+//                        + g_ent[i] * (-p_a * (log p_a + H_i))
+// ---------------------------------------------------------------------
+#define PG_MAX_A 32
+
+__global__ void pg_head_fwd_kernel(const __bf16* __restrict__ logits,
+                                   const int64_t* __restrict__ actions,
+                                   float* __restrict__ taken_logp,
+                                   float* __restrict__ entropy,
+                                   int64_t N, int A) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < N; i += (int64_t)gridDim.x * blockDim.x) {
+    float z[PG_MAX_A];
+    float zmax = -1e30f;
+    for (int a = 0; a < A; ++a) {
+      z[a] = (float)logits[i * A + a];
+      zmax = fmaxf(zmax, z[a]);
+    }
+    float sum = 0.0f;
+    for (int a = 0; a < A; ++a) {
+      z[a] = __expf(z[a] - zmax);
+      sum += z[a];
+    }
+    float inv = 1.0f / sum, logsum = __logf(sum);
+    float H = 0.0f;
+    for (int a = 0; a < A; ++a) {
+      float p = z[a] * inv;
+      float logp = __logf(z[a]) - logsum;  // = z_orig - zmax - logsum
+      H -= p * logp;
+    }
+    int ai = (int)actions[i];
+    taken_logp[i] = __logf(z[ai]) - logsum;
+    entropy[i] = H;
+  }
+}
+
+__global__ void pg_head_bwd_kernel(const __bf16* __restrict__ logits,
+                                   const int64_t* __restrict__ actions,
+                                   const float* __restrict__ g_taken,
+                                   const float* __restrict__ g_ent,
+                                   __bf16* __restrict__ dlogits,
+                                   int64_t N, int A) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < N; i += (int64_t)gridDim.x * blockDim.x) {
+    float z[PG_MAX_A];
+    float zmax = -1e30f;
+    for (int a = 0; a < A; ++a) {
+      z[a] = (float)logits[i * A + a];
+      zmax = fmaxf(zmax, z[a]);
+    }
+    float sum = 0.0f;
+    for (int a = 0; a < A; ++a) {
+      z[a] = __expf(z[a] - zmax);
+      sum += z[a];
+    }
+    float inv = 1.0f / sum, logsum = __logf(sum);
+    float H = 0.0f;
+    for (int a = 0; a < A; ++a) {
+      float p = z[a] * inv;
+      H -= p * (__logf(z[a]) - logsum);
+    }
+    int ai = (int)actions[i];
+    float gt = g_taken[i], ge = g_ent[i];
+    for (int a = 0; a < A; ++a) {
+      float p = z[a] * inv;
+      float logp = __logf(z[a]) - logsum;
+      float g = gt * ((a == ai ? 1.0f : 0.0f) - p)
+                - ge * p * (logp + H);
+      dlogits[i * A + a] = (__bf16)g;
+    }
+  }
+}
+
+void pg_head_fwd_launch(const void* logits, const int64_t* actions,
+                        float* taken_logp, float* entropy, int64_t N,
+                        int A, hipStream_t stream) {
+  const int block = 256;
+  hipLaunchKernelGGL(pg_head_fwd_kernel, dim3(ma_grid(N, block)),
+                     dim3(block), 0, stream, (const __bf16*)logits,
+                     actions, taken_logp, entropy, N, A);
+}
+
+void pg_head_bwd_launch(const void* logits, const int64_t* actions,
+                        const float* g_taken, const float* g_ent,
+                        void* dlogits, int64_t N, int A,
+                        hipStream_t stream) {
+  const int block = 256;
+  hipLaunchKernelGGL(pg_head_bwd_kernel, dim3(ma_grid(N, block)),
+                     dim3(block), 0, stream, (const __bf16*)logits,
+                     actions, g_taken, g_ent, (__bf16*)dlogits, N, A);
+}

@@ -213,6 +213,31 @@ class TestDistributionsGPU:
         ).sum(dim=1, keepdim=True)
         assert t.allclose(logp, expect, atol=5e-3, rtol=1e-3)
 
+    def test_categorical_policy_head_matches_torch(self, dev):
+        """Fused pg head: forward values AND analytic backward equal
+        the eager log_softmax/gather/entropy chain."""
+        import machin_amd.ops as ops
+
+        t.manual_seed(11)
+        N, A = 512, 6
+        logits = (t.randn(N, A) * 2).to(dev).to(t.bfloat16)
+        logits.requires_grad_(True)
+        actions = t.randint(0, A, (N,), device=dev)
+        w1 = t.randn(N, device=dev)
+        w2 = t.randn(N, device=dev)
+        tl, ent = ops.categorical_policy_head(logits, actions)
+        ((tl * w1).sum() + (ent * w2).sum()).backward()
+        g = logits.grad.clone().float()
+
+        ref = logits.detach().clone().requires_grad_(True)
+        logp = t.log_softmax(ref.float(), dim=-1)
+        tl2 = logp.gather(1, actions.view(-1, 1)).view(-1)
+        ent2 = -(logp.exp() * logp).sum(-1)
+        assert t.allclose(tl, tl2, rtol=1e-2, atol=2e-3)
+        assert t.allclose(ent, ent2, rtol=1e-2, atol=2e-3)
+        ((tl2 * w1).sum() + (ent2 * w2).sum()).backward()
+        assert t.allclose(g, ref.grad.float(), rtol=5e-2, atol=5e-3)
+
     def test_fused_head_gradients_match_torch(self, dev):
         """The analytic backward of the fused tanh-Gaussian head must
         equal autograd through the equivalent torch math (same

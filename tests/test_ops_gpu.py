@@ -213,6 +213,25 @@ class TestDistributionsGPU:
         ).sum(dim=1, keepdim=True)
         assert t.allclose(logp, expect, atol=5e-3, rtol=1e-3)
 
+    def test_soft_update_uses_cached_plan(self, dev):
+        """soft_update caches a FusedPolyak plan on the target net and
+        produces the polyak result (default GPU path)."""
+        import torch.nn as nn
+
+        from machin_amd.frame.algorithms.utils import soft_update
+
+        a = nn.Linear(16, 8).to(dev)
+        b = nn.Linear(16, 8).to(dev)
+        expect = [
+            pa.detach() * 0.99 + pb.detach() * 0.01
+            for pa, pb in zip(a.parameters(), b.parameters())
+        ]
+        soft_update(a, b, 0.01)
+        assert "_machin_polyak_plan" in a.__dict__
+        for p, e in zip(a.parameters(), expect):
+            assert t.allclose(p.detach(), e, atol=1e-6)
+        soft_update(a, b, 0.01)  # second call reuses the plan
+
     def test_categorical_policy_head_matches_torch(self, dev):
         """Fused pg head: forward values AND analytic backward equal
         the eager log_softmax/gather/entropy chain."""

@@ -500,7 +500,7 @@ def main():
     parser.add_argument("--bucket-mb", type=float, default=32.0,
                         help="GradReducer bucket size (MiB)")
     parser.add_argument("--ddp-reduction", default="all_reduce",
-                        choices=["all_reduce", "reduce_scatter"])
+                        choices=["all_reduce", "reduce_scatter", "one_shot"])
     parser.add_argument("--no-fused-stem", action="store_true",
                         help="fall back to MIOpen for the stem conv")
     args = parser.parse_args()

@@ -18,6 +18,12 @@ explicit reducer designed for one node of 8×MI355X:
   reduce-scatter + all-gather pair (half the per-link bytes of a ring
   all-reduce on xGMI for large buckets); falls back to all-reduce on
   backends without reduce_scatter support (gloo);
+* ``reduction="one_shot"`` all-gathers every rank's bucket and sums
+  locally — ONE xGMI hop instead of the ring's 2(N-1)/N round trips
+  (every peer is one hop away on this topology, so for
+  the reference's SMALL models — CartPole MLPs are a few KB — latency
+  dominates and the single-shot direct reduce wins; for large buckets
+  it moves N× the bytes, so keep all_reduce/reduce_scatter there);
 * no autograd graph rewriting, no "find_unused_parameters" machinery.
 
 Integration contract: after ``backward()`` call :meth:`finalize` (or
@@ -45,9 +51,11 @@ class GradReducer:
         average: bool = True,
         reduction: str = "all_reduce",
     ):
-        if reduction not in ("all_reduce", "reduce_scatter"):
+        if reduction not in ("all_reduce", "reduce_scatter",
+                             "one_shot"):
             raise ValueError(
-                "reduction must be 'all_reduce' or 'reduce_scatter'."
+                "reduction must be 'all_reduce', 'reduce_scatter' or "
+                "'one_shot'."
             )
         self.module = module
         self.group = process_group
@@ -144,6 +152,17 @@ class GradReducer:
                         group=self.group, async_op=True,
                     )
                     self._works.append(("rs", bi, work))
+                elif self.reduction == "one_shot":
+                    if "gathered" not in bucket:
+                        bucket["gathered"] = [
+                            t.empty_like(bucket["flat"])
+                            for _ in range(self.world_size)
+                        ]
+                    work = dist.all_gather(
+                        bucket["gathered"], bucket["flat"],
+                        group=self.group, async_op=True,
+                    )
+                    self._works.append(("ag", bi, work))
                 else:
                     work = dist.all_reduce(
                         bucket["flat"], op=dist.ReduceOp.SUM,
@@ -162,9 +181,14 @@ class GradReducer:
         gather_buckets = []
         for kind, bi, work in self._works:
             work.wait()
-            reduced_flats.append(self.buckets[bi]["flat"])
+            bucket = self.buckets[bi]
+            reduced_flats.append(bucket["flat"])
             if kind == "rs":
                 gather_buckets.append(bi)
+            elif kind == "ag":
+                # one-shot: local sum of the gathered per-rank buckets
+                t.sum(t.stack(bucket["gathered"]), dim=0,
+                      out=bucket["flat"])
         self._works.clear()
         if gather_buckets:
             gathers = [

@@ -164,6 +164,32 @@ class TestGradReducerRobustness:
 
         assert all(run_multi(fn, world_size=2))
 
+    def test_one_shot_mode(self):
+        """one_shot (single-hop all-gather + local sum)
+        produces the same averaged gradients as all_reduce."""
+        def fn(rank, world):
+            import torch.nn as nn
+
+            from machin_amd.parallel.ddp import GradReducer
+
+            t.manual_seed(0)
+            m1 = nn.Sequential(nn.Linear(8, 16), nn.Linear(16, 3))
+            t.manual_seed(0)
+            m2 = nn.Sequential(nn.Linear(8, 16), nn.Linear(16, 3))
+            r1 = GradReducer(m1, reduction="one_shot")
+            r2 = GradReducer(m2, reduction="all_reduce")
+            t.manual_seed(400 + rank)
+            x = t.rand(4, 8)
+            for r, m in ((r1, m1), (r2, m2)):
+                r.zero_grad_()
+                m(x).sum().backward()
+                r.finalize()
+            for p, q in zip(m1.parameters(), m2.parameters()):
+                assert t.allclose(p.grad, q.grad, atol=1e-6)
+            return True
+
+        assert all(run_multi(fn, world_size=2))
+
     def test_matches_torch_ddp(self):
         """GradReducer's reduced grads equal torch-DDP's on the same
         model + per-rank data (round-1 VERDICT next #2)."""

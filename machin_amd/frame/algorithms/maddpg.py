@@ -227,6 +227,30 @@ class MADDPG(TorchFramework):
         self.criterion = (
             criterion() if isinstance(criterion, type) else criterion
         )
+        # TorchScript actors: scripted modules SHARE parameters with
+        # the originals (updates through the optimizers are visible),
+        # and scripted forwards drop the GIL so act() parallelizes
+        # across agents on a thread pool.
+        self.use_jit = use_jit
+        self._jit_actors = None
+        self._jit_actor_targets = None
+        self._act_pool = None
+        if use_jit:
+            try:
+                self._jit_actors = [
+                    [t.jit.script(p) for p in agent_pols]
+                    for agent_pols in self.actors
+                ]
+                self._jit_actor_targets = [
+                    [t.jit.script(p) for p in agent_pols]
+                    for agent_pols in self.actor_targets
+                ]
+            except Exception as e:  # noqa: BLE001 - clearer message
+                raise ValueError(
+                    "use_jit=True requires TorchScript-compatible "
+                    f"actor models: {e}"
+                ) from e
+            self._act_pool = ThreadPool(processes=min(n, 8))
         if pool_type not in ("thread", "process"):
             raise ValueError(
                 f"pool_type must be 'thread' or 'process', got "
@@ -302,9 +326,29 @@ class MADDPG(TorchFramework):
         src = self.actor_targets if use_target else self.actors
         return [random.choice(agent_pols) for agent_pols in src]
 
+    def _jit_policies(self, use_target: bool):
+        src = self._jit_actor_targets if use_target else self._jit_actors
+        return [random.choice(agent_pols) for agent_pols in src]
+
     def act(self, states: List[Dict[str, Any]], use_target: bool = False,
             **__):
-        """Returns a list of action tensors, one per agent."""
+        """Returns a list of action tensors, one per agent.
+
+        With ``use_jit`` the per-agent forwards run CONCURRENTLY on a
+        thread pool: TorchScript forwards release the GIL, so N agents
+        act in parallel on one device (reference
+        machin/frame/algorithms/maddpg.py:278-303; the scripted
+        modules share parameter memory with the trained actors)."""
+        if self.use_jit:
+            pols = self._jit_policies(use_target)
+
+            def _one(pol, st):
+                with t.no_grad():
+                    return safe_return(pol(**st))
+
+            return self._act_pool.starmap(
+                _one, list(zip(pols, states))
+            )
         pols = self._policies(use_target)
         with t.no_grad():
             return [

@@ -151,3 +151,32 @@ class TestMADDPGProcessPool:
             assert step == 4, f"Adam state not shared: step={step}"
         finally:
             m.pool.terminate()
+
+
+class TestMADDPGJit:
+    def test_jit_actors_share_parameters_and_act(self):
+        """use_jit=True: scripted actors act (GIL-free thread
+        parallelism) and share parameter storage with the trained
+        modules (reference maddpg.py:278-303)."""
+        t.manual_seed(0)
+        m = make_maddpg(use_jit=True)
+        try:
+            states = [{"state": t.rand(1, 3)} for _ in range(3)]
+            acts = m.act(states)
+            assert len(acts) == 3 and acts[0].shape == (1, 1)
+            m.act_with_noise(states, noise_param=(0.0, 0.1),
+                             mode="normal")
+            m.store_episodes(make_episodes())
+            m.update()
+            for p, q in zip(
+                m.actors[0][0].parameters(),
+                m._jit_actors[0][0].parameters(),
+            ):
+                assert p.data_ptr() == q.data_ptr()
+            # jit act result matches the eager actor under no noise
+            st = {"state": t.rand(1, 3)}
+            eager = m.actors[0][0](st["state"])
+            jit = m._jit_actors[0][0](st["state"])
+            assert t.allclose(eager, jit, atol=1e-6)
+        finally:
+            m.pool.terminate()

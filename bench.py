@@ -113,10 +113,21 @@ class RingPipeline:
         ]
         M = pool_segments
         self.M = M
+        # frames pool is stored NHWC per step so a training gather is
+        # a ZERO-COPY view (the [T, B] layout forced a 1.2 GB
+        # transpose per step on the critical path); the permute to
+        # NHWC happens during ingest on the side stream instead
         self.pool = {
             k: t.zeros((M, *shape), dtype=dtype, device=device)
             for k, (shape, dtype) in self.spec.items()
+            if k != "frames"
         }
+        self.pool["frames"] = t.zeros(
+            (M, unroll, 84, 84, frames), dtype=t.uint8, device=device
+        )
+        self._frames_stage = t.empty(
+            (64, unroll, frames, 84, 84), dtype=t.uint8, device=device
+        )
         self.pinned = self.ring.make_pinned_staging(64)
         self.registered = False
         self.side = t.cuda.Stream()
@@ -133,9 +144,24 @@ class RingPipeline:
         for p in self.actors:
             p.start()
         self.registered = self.ring.host_register()
-        # pre-fill the pool so the first batches sample real segments
-        while self.filled < min(self.env_batch, self.M):
+        # pre-fill the pool so the first batches sample real segments;
+        # bounded so a stalled actor farm degrades to a partial pool
+        # (or a clear error) instead of hanging the driver's run
+        deadline = time.monotonic() + 120.0
+        target = min(self.env_batch, self.M)
+        while self.filled < target:
             self.ingest_once(timeout=2.0)
+            if time.monotonic() > deadline:
+                if self.filled == 0:
+                    raise RuntimeError(
+                        "rollout actors produced nothing within 120s"
+                    )
+                print(
+                    f"[bench] pool prefill timed out at "
+                    f"{self.filled}/{target} segments; continuing",
+                    file=sys.stderr,
+                )
+                break
         self._ingester = threading.Thread(
             target=self._ingest_loop, daemon=True
         )
@@ -146,15 +172,32 @@ class RingPipeline:
         if not idx:
             return 0
         wp, M, n = self.write_pos, self.M, len(idx)
+        positions = [(wp + i) % M for i in range(n)]
         with t.cuda.stream(self.side):
             if self.registered:
-                positions = [(wp + i) % M for i in range(n)]
-                self.ring.upload_slots(idx, self.pool, positions)
+                small = {
+                    k: v for k, v in self.pool.items() if k != "frames"
+                }
+                self.ring.upload_slots(idx, small, positions)
+                # frames: direct-DMA into TCHW staging, then one
+                # permuted device copy into the NHWC pool rows
+                for j, (slot_id, pos) in enumerate(zip(idx, positions)):
+                    self._frames_stage[j].copy_(
+                        self.ring.data["frames"][slot_id],
+                        non_blocking=True,
+                    )
+                stage = self._frames_stage[:n]
+                pos_t = t.tensor(positions, device=self.device)
+                self.pool["frames"].index_copy_(
+                    0, pos_t, stage.permute(0, 1, 3, 4, 2).contiguous()
+                )
             else:
                 batch = self.ring.gather(
                     idx, self.device, pinned=self.pinned
                 )
                 for k, v in batch.items():
+                    if k == "frames":
+                        v = v.permute(0, 1, 3, 4, 2).contiguous()
                     end = wp + n
                     if end <= M:
                         self.pool[k][wp:end] = v
@@ -178,31 +221,31 @@ class RingPipeline:
             self.ingest_once()
 
     def sample(self, out=None):
-        """Gather a t-major training batch from the HBM pool: one
-        index_select plus ONE strided copy straight into the
-        channels-last layout the conv stem wants (the round-1 chain
-        permute -> reshape -> to(channels_last) made three full passes
-        over the 1.2 GB frame batch). With ``out`` the gather writes
-        into preallocated static buffers (hipGraph replay mode)."""
+        """Gather a BATCH-MAJOR training batch from the HBM pool. The
+        frames pool is NHWC segment-major, so the gathered [B, T, H,
+        W, C] block IS the channels-last [B*T, C, H, W] image — one
+        index_select, zero transpose copies (the t-major layout paid
+        a 1.2 GB strided copy per step). With ``out`` the gather
+        writes into preallocated static buffers (hipGraph mode)."""
         B, T = self.env_batch, self.unroll
         seg = t.randint(0, max(self.filled, 1), (B,), device=self.device)
         p = self.pool
-        fr = p["frames"].index_select(0, seg)  # [B, T, C, H, W]
-        C, H, W = fr.shape[2:]
         if out is None:
-            frames = t.empty(
-                (T * B, C, H, W), dtype=fr.dtype, device=self.device,
-                memory_format=t.channels_last,
-            )
-            out = {"frames": frames}
-            out["frames"].view(T, B, C, H, W).copy_(fr.transpose(0, 1))
+            fr = p["frames"].index_select(0, seg)  # [B, T, H, W, C]
+            H, W, C = fr.shape[2:]
+            out = {
+                "frames": fr.view(B * T, H, W, C).permute(0, 3, 1, 2)
+            }
             for k in ("actions", "behavior_logp", "rewards",
                       "terminals"):
-                out[k] = p[k].index_select(0, seg).t().contiguous()
+                out[k] = p[k].index_select(0, seg)  # [B, T]
             return out
-        out["frames"].view(T, B, C, H, W).copy_(fr.transpose(0, 1))
+        H, W, C = p["frames"].shape[2:]
+        out["frames"].view(B, T, C, H, W).permute(0, 1, 3, 4, 2).copy_(
+            p["frames"].index_select(0, seg)
+        )
         for k in ("actions", "behavior_logp", "rewards", "terminals"):
-            out[k].copy_(p[k].index_select(0, seg).t())
+            out[k].copy_(p[k].index_select(0, seg))
         return out
 
     def sample_prefetched(self):
@@ -230,17 +273,17 @@ class RingPipeline:
 
     def make_static_batch(self):
         B, T = self.env_batch, self.unroll
-        C, H, W = self.pool["frames"].shape[2:]
+        H, W, C = self.pool["frames"].shape[2:]
         return {
             "frames": t.empty(
-                (T * B, C, H, W), dtype=t.uint8, device=self.device,
+                (B * T, C, H, W), dtype=t.uint8, device=self.device,
                 memory_format=t.channels_last,
             ),
-            "actions": t.empty((T, B), dtype=t.long,
+            "actions": t.empty((B, T), dtype=t.long,
                                device=self.device),
-            "behavior_logp": t.empty((T, B), device=self.device),
-            "rewards": t.empty((T, B), device=self.device),
-            "terminals": t.empty((T, B), device=self.device),
+            "behavior_logp": t.empty((B, T), device=self.device),
+            "rewards": t.empty((B, T), device=self.device),
+            "terminals": t.empty((B, T), device=self.device),
         }
 
     def reset_env_steps(self):
@@ -348,17 +391,17 @@ class ImpalaLearnerBench:
                         ).to(self.device)
                         .to(memory_format=t.channels_last),
                         "actions": t.randint(
-                            0, action_num, (unroll, env_batch),
+                            0, action_num, (env_batch, unroll),
                             generator=g,
                         ).to(self.device),
                         "behavior_logp": (
-                            -t.rand(unroll, env_batch, generator=g) * 2.0
+                            -t.rand(env_batch, unroll, generator=g) * 2.0
                         ).to(self.device),
                         "rewards": t.rand(
-                            unroll, env_batch, generator=g
+                            env_batch, unroll, generator=g
                         ).to(self.device),
                         "terminals": (
-                            t.rand(unroll, env_batch, generator=g) > 0.98
+                            t.rand(env_batch, unroll, generator=g) > 0.98
                         ).float().to(self.device),
                     }
                 )
@@ -403,16 +446,18 @@ class ImpalaLearnerBench:
                 frames = data["frames"].to(self.dtype).mul_(1.0 / 255.0)
             with t.autocast(device_type="cuda", dtype=self.dtype):
                 logits, values = self.model(frames)
-        values = values.float().view(T, B)
+        # everything batch-major [B, T] (flat row = b*T + t): the
+        # frames gather is then a zero-copy view of the NHWC pool
+        values = values.float().view(B, T)
         # fused policy head: ONE kernel for log_softmax+gather+entropy
         # (and one analytic backward) instead of the eager chain
         tl_flat, ent_flat = self.ops.categorical_policy_head(
-            logits.reshape(T * B, -1), data["actions"].reshape(-1)
+            logits.reshape(B * T, -1), data["actions"].reshape(-1)
         )
-        taken_logp = tl_flat.view(T, B)
+        taken_logp = tl_flat.view(B, T)
 
         with t.no_grad():
-            bootstrap = values[-1].detach()
+            bootstrap = values[:, -1].detach().contiguous()
             vs, pg_adv = self.ops.vtrace(
                 data["behavior_logp"],
                 taken_logp.detach(),
@@ -421,6 +466,7 @@ class ImpalaLearnerBench:
                 bootstrap,
                 data["terminals"],
                 self.discount,
+                time_major=False,
             )
 
         pg_loss = -(pg_adv * taken_logp).sum() / B

@@ -283,12 +283,36 @@ def vtrace(
     rho_clip: float = 1.0,
     c_clip: float = 1.0,
     pg_rho_clip: float = 1.0,
+    time_major: bool = True,
 ):
-    """IMPALA V-trace targets; shapes [T, B].
+    """IMPALA V-trace targets; shapes [T, B] (or [B, T] with
+    ``time_major=False`` — the batch-major kernel lets segment-major
+    HBM pools feed the update with zero-copy gathers).
 
-    Returns (vs [T,B], pg_advantages [T,B]), both detached.
+    Returns (vs, pg_advantages) in the input layout, both detached.
     Reference semantics: machin/frame/algorithms/impala.py:317-371.
     """
+    if not time_major:
+        if rewards.is_cuda:
+            ext = _require_ext()
+            B, T = rewards.shape
+            vs, pg_adv = ext.vtrace_bt(
+                behavior_log_probs.detach().float().contiguous(),
+                target_log_probs.detach().float().contiguous(),
+                rewards.detach().float().contiguous(),
+                values.detach().float().contiguous(),
+                bootstrap_value.detach().float().contiguous(),
+                (1.0 - terminals.detach().float()).contiguous(),
+                float(gamma), float(rho_clip), float(c_clip),
+                float(pg_rho_clip),
+            )
+            return vs, pg_adv
+        vs, pg_adv = vtrace(
+            behavior_log_probs.t(), target_log_probs.t(), rewards.t(),
+            values.t(), bootstrap_value, terminals.t(), gamma,
+            rho_clip, c_clip, pg_rho_clip,
+        )
+        return vs.t().contiguous(), pg_adv.t().contiguous()
     T, B = rewards.shape
     blp = behavior_log_probs.detach().float().view(T, B)
     tlp = target_log_probs.detach().float().view(T, B)

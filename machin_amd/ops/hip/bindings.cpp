@@ -26,6 +26,10 @@ void vtrace_launch(const float*, const float*, const float*, const float*,
                    int64_t, float, float, float, float, hipStream_t);
 void nstep_returns_launch(const float*, const float*, float*, int64_t,
                           int64_t, float, int, hipStream_t);
+void vtrace_bt_launch(const float*, const float*, const float*,
+                      const float*, const float*, const float*, float*,
+                      float*, int64_t, int64_t, float, float, float,
+                      float, hipStream_t);
 void categorical_projection_launch(const float*, const float*, const float*,
                                    float*, int64_t, int64_t, float, float,
                                    float, hipStream_t);
@@ -146,6 +150,24 @@ std::vector<Tensor> vtrace(Tensor blp, Tensor tlp, Tensor rew, Tensor val,
                 vs.data_ptr<float>(), pg_adv.data_ptr<float>(), T, B,
                 (float)gamma, (float)rho_clip, (float)c_clip,
                 (float)pg_rho_clip, current_stream());
+  return {vs, pg_adv};
+}
+
+std::vector<Tensor> vtrace_bt(Tensor blp, Tensor tlp, Tensor rew,
+                              Tensor val, Tensor bootstrap, Tensor nd,
+                              double gamma, double rho_clip,
+                              double c_clip, double pg_rho_clip) {
+  check_f32_cuda(rew, "rewards");
+  const at::hip::OptionalHIPGuardMasqueradingAsCUDA guard(rew.device());
+  int64_t B = rew.size(0), T = rew.size(1);
+  Tensor vs = at::empty_like(rew);
+  Tensor pg_adv = at::empty_like(rew);
+  vtrace_bt_launch(blp.data_ptr<float>(), tlp.data_ptr<float>(),
+                   rew.data_ptr<float>(), val.data_ptr<float>(),
+                   bootstrap.data_ptr<float>(), nd.data_ptr<float>(),
+                   vs.data_ptr<float>(), pg_adv.data_ptr<float>(), T, B,
+                   (float)gamma, (float)rho_clip, (float)c_clip,
+                   (float)pg_rho_clip, current_stream());
   return {vs, pg_adv};
 }
 
@@ -422,6 +444,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("discounted_returns", &discounted_returns);
   m.def("gae", &gae);
   m.def("vtrace", &vtrace);
+  m.def("vtrace_bt", &vtrace_bt);
   m.def("categorical_projection", &categorical_projection);
   m.def("multi_tensor_polyak", &multi_tensor_polyak);
   m.def("multi_tensor_polyak_cached", &multi_tensor_polyak_cached);

@@ -142,3 +142,56 @@ void nstep_returns_launch(const float* rew, const float* alive, float* out,
                      dim3(block), 0, stream, rew, alive, out, T, B, gamma,
                      n);
 }
+
+// batch-major V-trace: identical math to vtrace_kernel over [B, T]
+// row-major inputs (each batch row contiguous). Lets the e2e pipeline
+// train straight from segment-major HBM pools with ZERO-copy frame
+// gathers (the [T, B] layout forced a 1.2 GB transpose per step).
+// Tiny tensors (5 x T*B floats, L2-resident), so the per-lane-
+// contiguous access pattern is immaterial.
+__global__ void vtrace_bt_kernel(const float* __restrict__ blp,
+                                 const float* __restrict__ tlp,
+                                 const float* __restrict__ rew,
+                                 const float* __restrict__ val,
+                                 const float* __restrict__ bootstrap,
+                                 const float* __restrict__ nd,
+                                 float* __restrict__ vs_out,
+                                 float* __restrict__ pg_adv, int64_t T,
+                                 int64_t B, float gamma, float rho_clip,
+                                 float c_clip, float pg_rho_clip) {
+  for (int64_t b = blockIdx.x * blockDim.x + threadIdx.x; b < B;
+       b += (int64_t)gridDim.x * blockDim.x) {
+    float acc = 0.0f;
+    float vs_next = bootstrap[b];
+    float v_next = bootstrap[b];
+    const int64_t row = b * T;
+    for (int64_t t = T - 1; t >= 0; --t) {
+      int64_t k = row + t;
+      float rho = __expf(tlp[k] - blp[k]);
+      float ndk = nd[k];
+      float vk = val[k];
+      float td = rew[k] + gamma * ndk * v_next - vk;
+      float delta = fminf(rho, rho_clip) * td;
+      acc = delta + gamma * ndk * fminf(rho, c_clip) * acc;
+      float vs_t = acc + vk;
+      pg_adv[k] =
+          fminf(rho, pg_rho_clip) * (rew[k] + gamma * ndk * vs_next - vk);
+      vs_out[k] = vs_t;
+      vs_next = vs_t;
+      v_next = vk;
+    }
+  }
+}
+
+void vtrace_bt_launch(const float* blp, const float* tlp, const float* rew,
+                      const float* val, const float* bootstrap,
+                      const float* nd, float* vs_out, float* pg_adv,
+                      int64_t T, int64_t B, float gamma, float rho_clip,
+                      float c_clip, float pg_rho_clip,
+                      hipStream_t stream) {
+  const int block = 256;
+  hipLaunchKernelGGL(vtrace_bt_kernel, dim3(ma_grid(B, block)),
+                     dim3(block), 0, stream, blp, tlp, rew, val,
+                     bootstrap, nd, vs_out, pg_adv, T, B, gamma,
+                     rho_clip, c_clip, pg_rho_clip);
+}

@@ -125,11 +125,14 @@ class RolloutRing:
         non_blocking: bool = True,
     ):
         """Direct DMA: copy each ready slot into its HBM pool position
-        (requires :meth:`host_register`; issue on a side stream)."""
+        (requires :meth:`host_register`; issue on a side stream).
+        Attributes missing from ``pool`` are skipped (callers may
+        route some attributes through their own staging)."""
         for slot_id, pos in zip(indices, positions):
             for k, buf in self.data.items():
-                pool[k][pos].copy_(buf[slot_id],
-                                   non_blocking=non_blocking)
+                if k in pool:
+                    pool[k][pos].copy_(buf[slot_id],
+                                       non_blocking=non_blocking)
 
 
 class EpisodeSegmentCodec:

@@ -105,6 +105,28 @@ class TestScansGPU:
                       0.99, 0.95)
         assert t.allclose(gpu.cpu(), cpu, rtol=1e-4, atol=1e-3)
 
+    def test_vtrace_batch_major(self, dev):
+        """vtrace(time_major=False) equals the [T,B] kernel."""
+        import machin_amd.ops as ops
+
+        t.manual_seed(5)
+        T, B = 20, 64
+        blp, tlp = t.randn(T, B), t.randn(T, B) * 0.1
+        rew, val = t.rand(T, B), t.rand(T, B)
+        boot = t.rand(B)
+        term = (t.rand(T, B) > 0.9).float()
+        vs1, pg1 = ops.vtrace(blp.to(dev), tlp.to(dev), rew.to(dev),
+                              val.to(dev), boot.to(dev), term.to(dev),
+                              0.99)
+        vs2, pg2 = ops.vtrace(
+            blp.t().contiguous().to(dev), tlp.t().contiguous().to(dev),
+            rew.t().contiguous().to(dev), val.t().contiguous().to(dev),
+            boot.to(dev), term.t().contiguous().to(dev), 0.99,
+            time_major=False,
+        )
+        assert t.allclose(vs1, vs2.t(), rtol=1e-4, atol=1e-4)
+        assert t.allclose(pg1, pg2.t(), rtol=1e-4, atol=1e-4)
+
     def test_nstep_returns(self, dev):
         import machin_amd.ops as ops
 

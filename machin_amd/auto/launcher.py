@@ -118,15 +118,22 @@ def _dist_worker(rank: int, world_size: int, port: int, config_data: dict):
     from ..parallel.distributed.world import World
     from .envs.classic_control import make_dataset
 
-    World(world_size=world_size, rank=rank, name=str(rank),
-          dist_backend="nccl" if t.cuda.is_available() else "gloo",
-          dist_timeout=1800.0)
+    world = World(world_size=world_size, rank=rank, name=str(rank),
+                  dist_backend="nccl" if t.cuda.is_available() else "gloo",
+                  dist_timeout=1800.0)
     config = Config(**config_data)
     launcher = DistributedLauncher(config, dataset_factory=make_dataset)
     launcher.fit()
     # keep this process's control-plane services alive until every
-    # member finishes (A3C/APEX peers keep calling them)
+    # member finishes (A3C/APEX peers keep calling them), then tear
+    # the world down EXPLICITLY: destroying the gloo process group
+    # from interpreter-exit GC intermittently aborts the child
+    # ("terminate called without an active exception")
     dist.barrier()
+    try:
+        world.stop()
+    except Exception:  # noqa: BLE001 - teardown best-effort
+        pass
 
 
 def launch_distributed(config):

@@ -134,6 +134,10 @@ def _dist_worker(rank: int, world_size: int, port: int, config_data: dict):
         world.stop()
     except Exception:  # noqa: BLE001 - teardown best-effort
         pass
+    # exit WITHOUT interpreter teardown: lingering comm threads
+    # (gloo/pg destructors running from GC at exit) intermittently
+    # abort the child after training already succeeded
+    os._exit(0)
 
 
 def launch_distributed(config):

@@ -372,6 +372,8 @@ class ImpalaLearnerBench:
             self.model.parameters(), lr=lr, alpha=0.99, eps=0.1,
             capturable=capturable,
         )
+        self.capturable = capturable
+        self._fused_opt = None
 
         # synthetic rollout pool for --learner-only mode (uint8 frames
         # like a real Atari actor feed)
@@ -485,8 +487,25 @@ class ImpalaLearnerBench:
         loss.backward()
         if self.reducer is not None:
             self.reducer.finalize()
-        nn.utils.clip_grad_norm_(self.model.parameters(), self.grad_clip)
-        self.optim.step()
+        # fused clip+RMSprop (2 launches, no host sync); the first
+        # step runs eagerly to materialize the optimizer state the
+        # plan's pointer table needs
+        if self._fused_opt is not None and self._fused_opt.matches(
+            self.optim
+        ):
+            self._fused_opt.step()
+        else:
+            nn.utils.clip_grad_norm_(
+                self.model.parameters(), self.grad_clip
+            )
+            self.optim.step()
+            if not self.capturable:
+                try:
+                    self._fused_opt = self.ops.FusedRMSprop(
+                        self.optim, max_norm=self.grad_clip
+                    )
+                except (ValueError, RuntimeError):
+                    self._fused_opt = None
         return loss.detach()
 
     def capture_graph(self, warmup_steps: int = 3):

@@ -80,6 +80,98 @@ class FusedPolyak:
         )
 
 
+class FusedRMSprop:
+    """Cached-plan fused gradient-clip + RMSprop step (two kernel
+    launches, zero host synchronization — the eager
+    ``clip_grad_norm_`` + foreach-RMSprop chain is ~10 launches with
+    the clip threshold bouncing through device scalars).
+
+    Build from a ``torch.optim.RMSprop`` whose state is materialized
+    (run one eager step first); semantics match momentum=0,
+    centered=False, weight_decay=0 plus ``clip_grad_norm_(max_norm)``
+    (pass ``max_norm<=0`` to skip clipping). Parameters, grads and
+    ``square_avg`` state are updated in place through a precomputed
+    device pointer table.
+    """
+
+    def __init__(self, optimizer: "t.optim.RMSprop",
+                 max_norm: float = 0.0):
+        ext = _require_ext()
+        group = optimizer.param_groups[0]
+        if (
+            group.get("momentum", 0) != 0
+            or group.get("centered", False)
+            or group.get("weight_decay", 0) != 0
+            or len(optimizer.param_groups) != 1
+        ):
+            raise ValueError(
+                "FusedRMSprop supports a single param group with "
+                "momentum=0, centered=False, weight_decay=0."
+            )
+        self.lr = float(group["lr"])
+        self.alpha = float(group["alpha"])
+        self.eps = float(group["eps"])
+        self.max_norm = float(max_norm)
+        params = [p for p in group["params"] if p.requires_grad]
+        trips = []
+        for p in params:
+            st = optimizer.state.get(p)
+            if st is None or "square_avg" not in st:
+                raise ValueError(
+                    "optimizer state not materialized; run one eager "
+                    "step first"
+                )
+            g, sq = p.grad, st["square_avg"]
+            for x in (p, g, sq):
+                if not (
+                    x is not None and x.is_cuda and x.is_contiguous()
+                    and x.dtype == t.float32
+                ):
+                    raise ValueError(
+                        "FusedRMSprop needs contiguous fp32 CUDA "
+                        "params/grads/state"
+                    )
+            trips.append((p.data, g, sq))
+        n = len(trips)
+        table = t.empty(4 * n + 1, dtype=t.int64)
+        total = 0
+        for i, (pd, g, sq) in enumerate(trips):
+            table[3 * i] = pd.data_ptr()
+            table[3 * i + 1] = g.data_ptr()
+            table[3 * i + 2] = sq.data_ptr()
+            table[3 * n + i] = total
+            total += pd.numel()
+        table[4 * n] = total
+        dev = trips[0][0].device
+        self.table = table.to(dev)
+        self.norm_buf = t.zeros(1, dtype=t.float32, device=dev)
+        self.n = n
+        self.total = total
+        self._sig = tuple(
+            int(x.data_ptr()) for tr in trips for x in tr
+        )
+        self._ext = ext
+
+    def matches(self, optimizer) -> bool:
+        group = optimizer.param_groups[0]
+        sig = []
+        for p in group["params"]:
+            if not p.requires_grad:
+                continue
+            st = optimizer.state.get(p)
+            if st is None or p.grad is None:
+                return False
+            sig += [int(p.data.data_ptr()), int(p.grad.data_ptr()),
+                    int(st["square_avg"].data_ptr())]
+        return tuple(sig) == self._sig
+
+    def step(self):
+        self._ext.fused_rmsprop_cached(
+            self.table, self.n, self.total, self.norm_buf,
+            self.max_norm, self.lr, self.alpha, self.eps,
+        )
+
+
 def _load_ext():
     global _ext, _ext_error
     if _ext is not None or _ext_error is not None:
@@ -448,4 +540,5 @@ __all__ = [
     "categorical_projection",
     "categorical_policy_head",
     "FusedPolyak",
+    "FusedRMSprop",
 ]

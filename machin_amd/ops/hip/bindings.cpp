@@ -35,6 +35,9 @@ void categorical_projection_launch(const float*, const float*, const float*,
                                    float, hipStream_t);
 void multi_tensor_polyak_launch(void*, const int64_t*, int64_t, int64_t,
                                 float, hipStream_t);
+void fused_rmsprop_launch(void*, const int64_t*, int64_t, int64_t,
+                          float*, float, float, float, float,
+                          hipStream_t);
 void gaussian_sample_logprob_launch(const float*, const float*, float*,
                                     float*, int64_t, int64_t, uint64_t,
                                     uint64_t, int, float, hipStream_t);
@@ -249,6 +252,26 @@ void multi_tensor_polyak_cached(Tensor dev_table, int64_t n_tensors,
                              (float)tau, current_stream());
 }
 
+// Cached-plan fused clip+RMSprop: dev_table = [p,g,sq]*n ptrs then
+// the n+1 prefix; norm_buf = one fp32 scratch.
+void fused_rmsprop_cached(Tensor dev_table, int64_t n_tensors,
+                          int64_t total, Tensor norm_buf,
+                          double max_norm, double lr, double alpha,
+                          double eps) {
+  TORCH_CHECK(dev_table.is_cuda() &&
+                  dev_table.scalar_type() == at::kLong &&
+                  dev_table.is_contiguous(),
+              "dev_table must be contiguous int64 CUDA");
+  check_f32_cuda(norm_buf, "norm_buf");
+  const at::hip::OptionalHIPGuardMasqueradingAsCUDA guard(
+      dev_table.device());
+  int64_t* d = dev_table.data_ptr<int64_t>();
+  fused_rmsprop_launch((void*)d, d + 3 * n_tensors, n_tensors, total,
+                       norm_buf.data_ptr<float>(), (float)max_norm,
+                       (float)lr, (float)alpha, (float)eps,
+                       current_stream());
+}
+
 // ------------------------------------------------------------------
 // distributions
 // ------------------------------------------------------------------
@@ -448,6 +471,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("categorical_projection", &categorical_projection);
   m.def("multi_tensor_polyak", &multi_tensor_polyak);
   m.def("multi_tensor_polyak_cached", &multi_tensor_polyak_cached);
+  m.def("fused_rmsprop_cached", &fused_rmsprop_cached);
   m.def("nstep_returns", &nstep_returns);
   m.def("gaussian_sample_logprob", &gaussian_sample_logprob);
   m.def("gaussian_logprob", &gaussian_logprob);

@@ -65,3 +65,118 @@ void multi_tensor_polyak_launch(void* entries, const int64_t* prefix,
                      0, stream, (MTEntry*)entries, prefix, n_tensors, total,
                      tau);
 }
+
+// ---------------------------------------------------------------------
+// fused gradient-clip + RMSprop step, zero host synchronization.
+//
+// Replaces the eager chain clip_grad_norm_ (multi-tensor norm +
+// clamp + scale) followed by foreach-RMSprop (~10 launches over the
+// parameter list, with the clip threshold round-tripping through
+// device scalars) with TWO launches:
+//   pass 1: global sum of grad squares -> norm_buf[0] (block
+//           partials via one atomicAdd each);
+//   pass 2: reads the total IN-KERNEL, computes
+//           scale = min(1, max_norm / (sqrt(total) + 1e-6)),
+//           then per element:
+//             g' = g * scale
+//             sq = alpha*sq + (1 - alpha)*g'^2
+//             p -= lr * g' / (sqrt(sq) + eps)
+// Table layout (device int64): [p_ptr, g_ptr, sq_ptr] * n, then the
+// n+1 exclusive prefix of element counts (built once by the python
+// plan, like FusedPolyak).
+// ---------------------------------------------------------------------
+struct RMSEntry {
+  float* p;
+  float* g;
+  float* sq;
+};
+
+__global__ void grad_sqsum_kernel(RMSEntry* __restrict__ entries,
+                                  const int64_t* __restrict__ prefix,
+                                  int64_t n_tensors, int64_t total,
+                                  float* __restrict__ norm_buf) {
+  float acc = 0.0f;
+  for (int64_t start = (int64_t)blockIdx.x * MT_CHUNK; start < total;
+       start += (int64_t)gridDim.x * MT_CHUNK) {
+    int64_t lo = 0, hi = n_tensors - 1;
+    while (lo < hi) {
+      int64_t mid = (lo + hi + 1) >> 1;
+      if (prefix[mid] <= start) lo = mid; else hi = mid - 1;
+    }
+    int64_t ti = lo;
+    int64_t chunk_end = min(start + (int64_t)MT_CHUNK, total);
+    int64_t i = start + threadIdx.x;
+    while (i < chunk_end) {
+      while (ti + 1 < n_tensors && prefix[ti + 1] <= i) ++ti;
+      float g = entries[ti].g[i - prefix[ti]];
+      acc += g * g;
+      i += MT_BLOCK;
+    }
+  }
+  // block reduction then one atomic
+  __shared__ float red[MT_BLOCK / MA_WAVE];
+  acc += __shfl_down(acc, 32, 64);
+  acc += __shfl_down(acc, 16, 64);
+  acc += __shfl_down(acc, 8, 64);
+  acc += __shfl_down(acc, 4, 64);
+  acc += __shfl_down(acc, 2, 64);
+  acc += __shfl_down(acc, 1, 64);
+  int wave = threadIdx.x / MA_WAVE, lane = threadIdx.x % MA_WAVE;
+  if (lane == 0) red[wave] = acc;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    float s = 0.0f;
+    for (int w = 0; w < MT_BLOCK / MA_WAVE; ++w) s += red[w];
+    atomicAdd(norm_buf, s);
+  }
+}
+
+__global__ void rmsprop_step_kernel(RMSEntry* __restrict__ entries,
+                                    const int64_t* __restrict__ prefix,
+                                    int64_t n_tensors, int64_t total,
+                                    const float* __restrict__ norm_buf,
+                                    float max_norm, float lr,
+                                    float alpha, float eps) {
+  float scale = 1.0f;
+  if (max_norm > 0.0f) {
+    float norm = sqrtf(norm_buf[0]);
+    scale = fminf(1.0f, max_norm / (norm + 1e-6f));
+  }
+  for (int64_t start = (int64_t)blockIdx.x * MT_CHUNK; start < total;
+       start += (int64_t)gridDim.x * MT_CHUNK) {
+    int64_t lo = 0, hi = n_tensors - 1;
+    while (lo < hi) {
+      int64_t mid = (lo + hi + 1) >> 1;
+      if (prefix[mid] <= start) lo = mid; else hi = mid - 1;
+    }
+    int64_t ti = lo;
+    int64_t chunk_end = min(start + (int64_t)MT_CHUNK, total);
+    int64_t i = start + threadIdx.x;
+    while (i < chunk_end) {
+      while (ti + 1 < n_tensors && prefix[ti + 1] <= i) ++ti;
+      int64_t off = i - prefix[ti];
+      float g = entries[ti].g[off] * scale;
+      float sq = alpha * entries[ti].sq[off] + (1.0f - alpha) * g * g;
+      entries[ti].sq[off] = sq;
+      entries[ti].p[off] -= lr * g / (sqrtf(sq) + eps);
+      i += MT_BLOCK;
+    }
+  }
+}
+
+void fused_rmsprop_launch(void* entries, const int64_t* prefix,
+                          int64_t n_tensors, int64_t total,
+                          float* norm_buf, float max_norm, float lr,
+                          float alpha, float eps, hipStream_t stream) {
+  if (total == 0) return;
+  HIP_CHECK(hipMemsetAsync(norm_buf, 0, sizeof(float), stream));
+  int grid = ma_grid((total + MT_CHUNK - 1) / MT_CHUNK, 1);
+  if (max_norm > 0.0f) {
+    hipLaunchKernelGGL(grad_sqsum_kernel, dim3(grid), dim3(MT_BLOCK), 0,
+                       stream, (RMSEntry*)entries, prefix, n_tensors,
+                       total, norm_buf);
+  }
+  hipLaunchKernelGGL(rmsprop_step_kernel, dim3(grid), dim3(MT_BLOCK), 0,
+                     stream, (RMSEntry*)entries, prefix, n_tensors,
+                     total, norm_buf, max_norm, lr, alpha, eps);
+}

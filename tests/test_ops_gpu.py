@@ -235,6 +235,49 @@ class TestDistributionsGPU:
         ).sum(dim=1, keepdim=True)
         assert t.allclose(logp, expect, atol=5e-3, rtol=1e-3)
 
+    def test_fused_rmsprop_matches_torch(self, dev):
+        """Fused clip+RMSprop equals eager clip_grad_norm_ +
+        torch.optim.RMSprop over multiple steps."""
+        import copy
+
+        import torch.nn as nn
+
+        import machin_amd.ops as ops
+
+        t.manual_seed(9)
+        m1 = nn.Sequential(nn.Linear(16, 32), nn.ReLU(),
+                           nn.Linear(32, 4)).to(dev)
+        m2 = copy.deepcopy(m1)
+        o1 = t.optim.RMSprop(m1.parameters(), lr=1e-2, alpha=0.99,
+                             eps=0.1)
+        o2 = t.optim.RMSprop(m2.parameters(), lr=1e-2, alpha=0.99,
+                             eps=0.1)
+        clip = 0.5  # small enough that clipping actually engages
+        plan = None
+        for step in range(6):
+            t.manual_seed(100 + step)
+            x = t.rand(8, 16, device=dev)
+            for m, o in ((m1, o1), (m2, o2)):
+                o.zero_grad(set_to_none=False)
+                (m(x) ** 2).sum().backward()
+            # reference: eager
+            nn.utils.clip_grad_norm_(m1.parameters(), clip)
+            o1.step()
+            # fused (first step eager to materialize state)
+            if plan is None:
+                nn.utils.clip_grad_norm_(m2.parameters(), clip)
+                o2.step()
+                plan = ops.FusedRMSprop(o2, max_norm=clip)
+            else:
+                assert plan.matches(o2)
+                plan.step()
+        for p1, p2 in zip(m1.parameters(), m2.parameters()):
+            assert t.allclose(p1, p2, rtol=1e-4, atol=1e-5)
+        for p1, p2 in zip(m1.parameters(), m2.parameters()):
+            s1 = o1.state[p1]["square_avg"]
+            s2 = o2.state[p2]["square_avg"]
+            assert t.allclose(s1, s2, rtol=1e-4, atol=1e-6)
+
     def test_soft_update_uses_cached_plan(self, dev):
         """soft_update caches a FusedPolyak plan on the target net and
         produces the polyak result (default GPU path)."""

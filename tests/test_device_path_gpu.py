@@ -259,3 +259,43 @@ class TestIMPALARingGPU:
             return True
 
         assert all(run_multi(fn, world_size=1, timeout=300))
+
+
+class TestTD3DeviceReplay:
+    def test_td3_trains_from_hbm(self):
+        """TD3 (DDPG family) with replay_device=cuda: twin critics,
+        target smoothing, HBM ring replay."""
+        from machin_amd.auto.model_zoo import (
+            DeterministicActor,
+            QCritic,
+        )
+        from machin_amd.frame.algorithms import TD3
+        from machin_amd.frame.buffers.device_buffer import (
+            DeviceTransitionBuffer,
+        )
+
+        dev = "cuda:0"
+        frame = TD3(
+            DeterministicActor(3, 1).to(dev),
+            DeterministicActor(3, 1).to(dev),
+            QCritic(3, 1).to(dev), QCritic(3, 1).to(dev),
+            QCritic(3, 1).to(dev), QCritic(3, 1).to(dev),
+            t.optim.Adam, nn.MSELoss(),
+            replay_device=dev, batch_size=16,
+        )
+        assert isinstance(frame.replay_buffer, DeviceTransitionBuffer)
+        for _ in range(4):
+            ep = [
+                {
+                    "state": {"state": t.rand(1, 3)},
+                    "action": {"action": t.rand(1, 1) * 2 - 1},
+                    "next_state": {"state": t.rand(1, 3)},
+                    "reward": float(t.rand(1)),
+                    "terminal": i == 4,
+                }
+                for i in range(5)
+            ]
+            frame.store_episode(ep)
+        for _ in range(4):
+            out = frame.update()
+            assert all(v == v for v in out)

@@ -182,7 +182,8 @@ class BasePool:
                 self._workers[i] = new
 
     # -- submission ----------------------------------------------------
-    def _submit(self, func, args, kwargs, needs_ctx=False) -> AsyncResult:
+    def _submit(self, func, args, kwargs, needs_ctx=False,
+                _func_bytes=None) -> AsyncResult:
         if self._closed:
             raise RuntimeError("Pool is closed.")
         self._maintain_workers()
@@ -193,13 +194,20 @@ class BasePool:
         self._put_task(
             (
                 task_id,
-                dumps(func, recurse=True, copy_tensor=self.copy_tensor),
+                _func_bytes if _func_bytes is not None else dumps(
+                    func, recurse=True, copy_tensor=self.copy_tensor
+                ),
                 args,
                 kwargs,
                 needs_ctx,
             )
         )
         return res
+
+    def _serialize_func(self, func) -> bytes:
+        """Serialize a task function ONCE for a whole map/starmap
+        call (dill per task dominated small-task dispatch)."""
+        return dumps(func, recurse=True, copy_tensor=self.copy_tensor)
 
     def _put_task(self, task):
         self._task_queue.put(task)
@@ -219,25 +227,33 @@ class BasePool:
 
     def map_async(self, func, iterable, chunksize=None) -> AsyncResult:
         items = list(iterable)
+        fb = self._serialize_func(func)
         return self._gather_async(
-            [self._submit(func, (x,), {}) for x in items]
+            [self._submit(func, (x,), {}, _func_bytes=fb)
+             for x in items]
         )
 
     def starmap(self, func, iterable, chunksize=None) -> list:
         return self.starmap_async(func, iterable, chunksize).get()
 
     def starmap_async(self, func, iterable, chunksize=None) -> AsyncResult:
+        fb = self._serialize_func(func)
         return self._gather_async(
-            [self._submit(func, tuple(args), {}) for args in iterable]
+            [self._submit(func, tuple(args), {}, _func_bytes=fb)
+             for args in iterable]
         )
 
     def imap(self, func, iterable, chunksize=None):
-        results = [self._submit(func, (x,), {}) for x in iterable]
+        fb = self._serialize_func(func)
+        results = [self._submit(func, (x,), {}, _func_bytes=fb)
+                   for x in iterable]
         for r in results:
             yield r.get()
 
     def imap_unordered(self, func, iterable, chunksize=None):
-        results = [self._submit(func, (x,), {}) for x in iterable]
+        fb = self._serialize_func(func)
+        results = [self._submit(func, (x,), {}, _func_bytes=fb)
+                   for x in iterable]
         pending = set(results)
         while pending:
             for r in list(pending):
@@ -369,12 +385,17 @@ class CtxPool(BasePool):
             copy_tensor=copy_tensor,
         )
 
-    def _submit(self, func, args, kwargs, needs_ctx=True):
-        return super()._submit(func, args, kwargs, needs_ctx=True)
+    def _submit(self, func, args, kwargs, needs_ctx=True,
+                _func_bytes=None):
+        return super()._submit(func, args, kwargs, needs_ctx=True,
+                               _func_bytes=_func_bytes)
 
 
 class ThreadPool:
     """Thread pool with the same call surface (no serialization)."""
+
+    def _serialize_func(self, func):
+        return None  # threads share memory; nothing to serialize
 
     def __init__(self, processes: int = None, initializer=None,
                  initargs=(), worker_contexts=None, **__):
@@ -411,7 +432,8 @@ class ThreadPool:
             except Exception as e:  # noqa: BLE001
                 res._set(False, (repr(e), traceback.format_exc()))
 
-    def _submit(self, func, args, kwargs, needs_ctx=False) -> AsyncResult:
+    def _submit(self, func, args, kwargs, needs_ctx=False,
+                _func_bytes=None) -> AsyncResult:
         if self._closed:
             raise RuntimeError("Pool is closed.")
         res = AsyncResult()
@@ -463,5 +485,7 @@ class CtxThreadPool(ThreadPool):
             worker_contexts=worker_contexts,
         )
 
-    def _submit(self, func, args, kwargs, needs_ctx=True):
-        return super()._submit(func, args, kwargs, needs_ctx=True)
+    def _submit(self, func, args, kwargs, needs_ctx=True,
+                _func_bytes=None):
+        return super()._submit(func, args, kwargs, needs_ctx=True,
+                               _func_bytes=_func_bytes)

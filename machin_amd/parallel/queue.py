@@ -32,9 +32,16 @@ class SimpleQueue:
 
     def put(self, obj: Any, timeout: float = None):
         data = dumps(obj, copy_tensor=self.copy_tensor)
-        if not self._wlock.acquire(
-            timeout=timeout if timeout is not None else -1
-        ):
+        # NOTE: multiprocessing SemLock treats a NEGATIVE timeout as
+        # try-once, not infinite (unlike threading.Lock) — passing -1
+        # made contended put/get kill pool workers with spurious
+        # Full/Empty under load
+        acquired = (
+            self._wlock.acquire()
+            if timeout is None
+            else self._wlock.acquire(timeout=timeout)
+        )
+        if not acquired:
             raise Full("put timed out")
         try:
             self._writer.send_bytes(data)
@@ -43,9 +50,12 @@ class SimpleQueue:
 
     def get(self, timeout: float = None) -> Any:
         deadline = None if timeout is None else time.monotonic() + timeout
-        if not self._rlock.acquire(
-            timeout=timeout if timeout is not None else -1
-        ):
+        acquired = (
+            self._rlock.acquire()
+            if timeout is None
+            else self._rlock.acquire(timeout=timeout)
+        )
+        if not acquired:
             raise Empty("get timed out")
         try:
             remain = (
@@ -131,15 +141,23 @@ class MultiP2PQueue:
         self._put_idx = (self._put_idx + 1) % len(self.queues)
 
     def get(self, timeout: float = None) -> Any:
-        """Get from any sub-queue (poll loop)."""
+        """Get from any sub-queue — event-driven multiplexing over
+        the pipe readers (the previous 100 µs sleep-poll loop capped
+        small-result throughput)."""
+        from multiprocessing.connection import wait
+
         deadline = None if timeout is None else time.monotonic() + timeout
+        readers = {q._reader: q for q in self.queues}
         while True:
-            for q in self.queues:
-                if not q.empty():
-                    return q.get()
-            if deadline is not None and time.monotonic() > deadline:
+            remain = (
+                None if deadline is None
+                else max(deadline - time.monotonic(), 0)
+            )
+            ready = wait(list(readers), timeout=remain)
+            if ready:
+                return readers[ready[0]].get()
+            if deadline is not None and time.monotonic() >= deadline:
                 raise Empty("get timed out")
-            time.sleep(1e-4)
 
     def get_sub_queue(self, index: int) -> SimpleP2PQueue:
         return self.queues[index]

@@ -225,3 +225,67 @@ class TestMultiP2PQueue:
         with pytest.raises(Empty):
             q.get(timeout=0.05)
         q.close()
+
+
+def _inc(x):
+    return x + 1
+
+
+class TestQueueContention:
+    def test_simple_queue_contended_get_put(self):
+        """Regression: multiprocessing SemLocks treat negative
+        timeouts as TRY-ONCE; SimpleQueue used timeout=-1 for
+        "blocking" acquire, so contended put/get raised spurious
+        Full/Empty and killed pool workers under load."""
+        import threading
+
+        from machin_amd.parallel.queue import SimpleQueue
+
+        q = SimpleQueue()
+        n_threads, per_thread = 8, 200
+        errors = []
+
+        def producer(base):
+            try:
+                for i in range(per_thread):
+                    q.put(base + i)
+            except Exception as e:  # noqa: BLE001
+                errors.append(e)
+
+        got = []
+        got_lock = threading.Lock()
+
+        def consumer():
+            try:
+                for _ in range(per_thread * n_threads // 4):
+                    v = q.get(timeout=30)
+                    with got_lock:
+                        got.append(v)
+            except Exception as e:  # noqa: BLE001
+                errors.append(e)
+
+        producers = [
+            threading.Thread(target=producer, args=(k * per_thread,))
+            for k in range(n_threads)
+        ]
+        consumers = [threading.Thread(target=consumer) for _ in range(4)]
+        for th in producers + consumers:
+            th.start()
+        for th in producers + consumers:
+            th.join(timeout=60)
+        assert not errors, errors
+        assert len(got) == n_threads * per_thread
+        assert len(set(got)) == len(got)
+
+    def test_pool_survives_many_small_tasks(self):
+        """2000 chunk-1 tasks through the process pool: no worker
+        churn, all results correct (previously workers died on
+        contended queue locks)."""
+        from machin_amd.parallel.pool import Pool
+
+        p = Pool(processes=4)
+        try:
+            out = p.map(_inc, range(500))
+            assert out == [x + 1 for x in range(500)]
+        finally:
+            p.terminate()

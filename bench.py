@@ -558,8 +558,8 @@ def main():
                         help="skip the actor farm/ring; measure the "
                              "bare learner loop from resident pools")
     parser.add_argument("--graph", action="store_true",
-                        help="force hipGraph capture of the learner "
-                             "step (default: on for single-GPU runs)")
+                        help="capture the learner step in a hipGraph "
+                             "(opt-in; measured net-negative here)")
     parser.add_argument("--no-graph", action="store_true",
                         help="disable hipGraph capture")
     parser.add_argument("--bucket-mb", type=float, default=32.0,

@@ -232,3 +232,26 @@ class TestPeerLiveness:
             return all(status.values()) and len(status) == 3
 
         assert all(run_multi(fn))
+
+
+class TestAsyncExecutorBurst:
+    def test_many_concurrent_async_calls(self):
+        """100 registered_async calls in flight complete on the shared
+        executor (round-1 weak #9: thread-per-call churned at APEX
+        fan-out rates)."""
+        def fn(rank, world):
+            group = world.create_rpc_group("burst", ["0", "1", "2"])
+            group.register(f"echo_{rank}", lambda x: x * 2)
+            group.barrier()
+            futures = [
+                group.registered_async(
+                    f"echo_{(rank + 1) % 3}", args=(i,)
+                )
+                for i in range(100)
+            ]
+            results = [f.wait() for f in futures]
+            assert results == [i * 2 for i in range(100)]
+            group.barrier()
+            return True
+
+        assert all(run_multi(fn))

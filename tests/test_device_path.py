@@ -209,3 +209,50 @@ class TestApexDeviceBuffer:
 
         results = run_multi(fn, world_size=3, timeout=240)
         assert results[0] and results[1]
+
+
+class TestDeviceBufferWraparound:
+    def test_ring_eviction_and_sampling_after_wrap(self):
+        from machin_amd.frame.buffers import DeviceTransitionBuffer
+
+        buf = DeviceTransitionBuffer(12, "cpu", prioritized=True)
+        for k in range(5):  # 25 transitions through a 12-slot ring
+            ep = [
+                {
+                    "state": {"state": t.full((1, 4), float(k * 5 + i))},
+                    "action": {"action": t.randint(0, 2, (1, 1))},
+                    "next_state": {"state": t.rand(1, 4)},
+                    "reward": float(k * 5 + i),
+                    "terminal": i == 4,
+                }
+                for i in range(5)
+            ]
+            buf.store_episode(ep)
+        assert buf.size() == 12  # clamped to capacity
+        bs, batch, idx, w = buf.sample_batch(
+            64,
+            sample_attrs=["state", "action", "reward", "next_state",
+                          "terminal", "*"],
+        )
+        # only the newest 12 rewards (13..24) can be sampled
+        rewards = batch[2].view(-1)
+        assert rewards.min() >= 13.0 and rewards.max() <= 24.0
+        # ring positions stay in range after wrap
+        assert int(idx.max()) < 16  # capacity rounded to tree size
+        buf.update_priority(t.rand(64), idx)
+
+    def test_capacity_16_exact_positions(self):
+        from machin_amd.frame.buffers.device_buffer import (
+            DeviceReplayBuffer,
+        )
+
+        buf = DeviceReplayBuffer(
+            8, {"x": ((2,), t.float32)}, "cpu"
+        )
+        pos1 = buf.store_batch({"x": t.arange(12.0).view(6, 2)})
+        assert pos1.tolist() == [0, 1, 2, 3, 4, 5]
+        pos2 = buf.store_batch({"x": t.arange(8.0).view(4, 2)})
+        assert pos2.tolist() == [6, 7, 0, 1]  # wrapped
+        assert buf.size() == 8
+        # wrapped rows hold the new data
+        assert buf.data["x"][0].tolist() == [4.0, 5.0]

@@ -193,3 +193,20 @@ class TestPixelCatch:
         frame.store_episode(episode)
         loss = frame.update()
         assert np.isfinite(loss)
+
+
+class TestPixelCatchSubProc:
+    def test_pixel_env_through_subprocess_pool(self):
+        """PixelCatch frames (84x84x4 u8) round-trip the subprocess
+        env pool (dill creators, shared result queue)."""
+        wrapper = ParallelWrapperSubProc(
+            [lambda *_: PixelCatchEnv(seed=0, balls=1)] * 2
+        )
+        try:
+            obs = wrapper.reset()
+            assert len(obs) == 2
+            assert obs[0].shape == (4, 84, 84)
+            obs, rew, done, info = wrapper.step([1, 2])
+            assert obs[1].shape == (4, 84, 84)
+        finally:
+            wrapper.close()

@@ -75,6 +75,54 @@ class TestDQNApexMultiLearner:
         assert results[2] is None
 
 
+class TestDQNApexOneShotReduction:
+    def test_one_shot_config_path(self):
+        """ddp_reduction="one_shot" flows from the config into the
+        learner GradReducer and keeps ranks in sync."""
+        def fn(rank, world):
+            from machin_amd.frame.algorithms import DQNApex
+
+            config = DQNApex.generate_config({})
+            fc = config["frame_config"]
+            fc["models"] = [_ZOO + "QNet", _ZOO + "QNet"]
+            fc["model_kwargs"] = (
+                {"state_dim": 4, "action_num": 2},
+                {"state_dim": 4, "action_num": 2},
+            )
+            fc["learner_process_number"] = 2
+            fc["ddp_reduction"] = "one_shot"
+            fc["batch_size"] = 8
+            fc["replay_size"] = 100
+            frame = DQNApex.init_from_config(config)
+            group = world.groups["apex_group"]
+            group.barrier()
+            if rank in (0, 1):
+                assert frame.qnet.reducer.reduction == "one_shot"
+            t.manual_seed(1500 + rank)
+            episode = [
+                {
+                    "state": {"state": t.rand(1, 4)},
+                    "action": {"action": t.randint(0, 2, (1, 1))},
+                    "next_state": {"state": t.rand(1, 4)},
+                    "reward": float(t.rand(1)),
+                    "terminal": i == 4,
+                }
+                for i in range(5)
+            ]
+            frame.store_episode(episode)
+            group.barrier()
+            out = None
+            if rank in (0, 1):
+                for _ in range(2):
+                    frame.update()
+                out = _flat_params(frame.qnet)
+            group.barrier()
+            return out
+
+        results = run_multi(fn, world_size=3, timeout=240)
+        assert np.allclose(results[0], results[1], atol=1e-6)
+
+
 class TestIMPALAMultiLearner:
     def test_two_learners_stay_in_sync(self):
         def fn(rank, world):

@@ -41,12 +41,21 @@ class StochasticActor(nn.Module):
 
     def forward(self, state, action=None):
         logits = self.fc2(t.relu(self.fc1(state)))
-        dist = t.distributions.Categorical(logits=logits)
         if action is None:
+            dist = t.distributions.Categorical(logits=logits)
             action = dist.sample().view(-1, 1)
-        log_prob = dist.log_prob(action.view(-1)).view(-1, 1)
-        entropy = dist.entropy().view(-1, 1)
-        return action, log_prob, entropy
+            log_prob = dist.log_prob(action.view(-1)).view(-1, 1)
+            entropy = dist.entropy().view(-1, 1)
+            return action, log_prob, entropy
+        # evaluation of GIVEN actions (the A2C/PPO/IMPALA update
+        # path): fused log_softmax+gather+entropy kernel on ROCm,
+        # identical torch math on CPU
+        from .. import ops
+
+        log_prob, entropy = ops.categorical_policy_head(
+            logits, action.view(-1)
+        )
+        return action, log_prob.view(-1, 1), entropy.view(-1, 1)
 
 
 class DeterministicActor(nn.Module):

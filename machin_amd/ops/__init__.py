@@ -507,9 +507,10 @@ def categorical_policy_head(logits: t.Tensor, actions: t.Tensor):
     logits (analytic softmax/entropy gradients in the backward
     kernel). CPU fallback uses the identical torch math.
     """
-    if logits.is_cuda:
-        if logits.dtype != t.bfloat16:
-            logits = logits.to(t.bfloat16)
+    if logits.is_cuda and logits.dtype == t.bfloat16:
+        # the fused kernel consumes bf16 logits (the bench/CNN path);
+        # fp32 logits keep the exact eager math below rather than
+        # silently quantizing to bf16
         return _PGHead.apply(logits, actions.long().view(-1))
     logp = t.log_softmax(logits.float(), dim=-1)
     tl = logp.gather(

@@ -39,12 +39,19 @@ def _share_optimizer_state(opt):
     every state tensor into shared memory so P2PPool workers' steps
     accumulate into the PARENT's Adam moments (a worker receives the
     optimizer by reference-serialization each call)."""
+    saved_wd = []
     for group in opt.param_groups:
+        # weight decay would turn the materialization step into a
+        # real parameter update even with zero gradients
+        saved_wd.append(group.get("weight_decay", 0))
+        group["weight_decay"] = 0
         for p in group["params"]:
             if p.grad is None:
                 p.grad = t.zeros_like(p)
-    opt.step()  # zero grad: params unchanged, state allocated
+    opt.step()  # zero grad + zero wd: params unchanged, state built
     opt.zero_grad(set_to_none=False)
+    for group, wd in zip(opt.param_groups, saved_wd):
+        group["weight_decay"] = wd
     for st in opt.state.values():
         for v in st.values():
             if t.is_tensor(v):
